@@ -1,0 +1,76 @@
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from video_features_amd.config import Config
+from video_features_amd.models.clip_vit import ViTConfig, VisionTransformer
+
+
+def tiny_vit():
+    return VisionTransformer(ViTConfig(input_resolution=64, patch_size=16,
+                                       width=64, layers=2, heads=4,
+                                       output_dim=32))
+
+
+def test_vit_shapes():
+    torch.manual_seed(0)
+    m = tiny_vit().eval()
+    x = torch.randn(3, 3, 64, 64)
+    with torch.no_grad():
+        out = m.encode_image(x)
+    assert out.shape == (3, 32)
+    assert torch.isfinite(out).all()
+
+
+def test_vit_attention_matches_sdpa():
+    # our ops.attention torch path vs torch's scaled_dot_product_attention
+    from video_features_amd import ops
+    torch.manual_seed(1)
+    q, k, v = [torch.randn(2, 4, 10, 16) for _ in range(3)]
+    ours = ops.attention(q, k, v)
+    ref = torch.nn.functional.scaled_dot_product_attention(q, k, v)
+    assert torch.allclose(ours, ref, atol=1e-5)
+
+
+def test_clip_extractor_end_to_end(tmp_path, y4m_video):
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[y4m_video],
+                 extract_method='uni_4', cpu=True, on_extraction='save_numpy',
+                 output_path=str(tmp_path / 'out'), tmp_path=str(tmp_path / 'tmp'))
+    from video_features_amd.extractors.clip import ExtractCLIP
+    ex = ExtractCLIP(cfg)
+    ex(torch.arange(1))
+    out_dir = os.path.join(str(tmp_path / 'out'), 'CLIP-ViT-B/32')
+    files = os.listdir(out_dir)
+    assert 'vid_CLIP-ViT-B_32.npy' in files
+    feats = np.load(os.path.join(out_dir, 'vid_CLIP-ViT-B_32.npy'))
+    assert feats.shape == (4, 512)
+    assert np.isfinite(feats).all()
+
+
+def test_clip_external_call(y4m_video):
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[y4m_video],
+                 extract_method='uni_3', cpu=True)
+    from video_features_amd.extractors.clip import ExtractCLIP
+    ex = ExtractCLIP(cfg, external_call=True)
+    feats_list = ex(torch.arange(1))
+    assert len(feats_list) == 1
+    fd = feats_list[0]
+    assert fd['CLIP-ViT-B/32'].shape == (3, 512)
+    assert 'fps' in fd and 'timestamps_ms' in fd
+    assert len(fd['timestamps_ms']) == 3
+
+
+def test_error_isolation(tmp_path, y4m_video, capsys):
+    # a corrupt video must not kill the shard (reference extract_clip.py:70-84)
+    bad = tmp_path / 'bad.y4m'
+    bad.write_bytes(b'garbage not a y4m file at all')
+    cfg = Config(feature_type='CLIP-ViT-B/32',
+                 video_paths=[str(bad), y4m_video],
+                 extract_method='uni_2', cpu=True)
+    from video_features_amd.extractors.clip import ExtractCLIP
+    ex = ExtractCLIP(cfg, external_call=True)
+    feats_list = ex(torch.arange(2))
+    assert len(feats_list) == 1      # good video still extracted
+    assert 'Extraction failed' in capsys.readouterr().out

@@ -1,0 +1,102 @@
+import argparse
+import os
+import pickle
+
+import numpy as np
+import pytest
+
+from video_features_amd.config import Config, sanity_check
+from video_features_amd.runtime.sinks import (action_on_extraction,
+                                              make_output_path, outputs_exist)
+
+
+def test_config_from_namespace():
+    ns = argparse.Namespace(feature_type='i3d', video_paths=['x.mp4'],
+                            flow_type='raft', unknown_flag=42)
+    cfg = Config.coerce(ns)
+    assert cfg.feature_type == 'i3d'
+    assert cfg.flow_type == 'raft'
+    assert cfg.on_extraction == 'print'   # default preserved
+
+
+def test_config_validation():
+    with pytest.raises(ValueError):
+        Config(feature_type='nope')
+    with pytest.raises(ValueError):
+        Config(on_extraction='save_hdf5')
+
+
+def test_sanity_check_rules():
+    sanity_check(Config(feature_type='i3d', stack_size=64))
+    with pytest.raises(ValueError):
+        sanity_check(Config(feature_type='i3d', stack_size=4))
+    with pytest.raises(ValueError):
+        sanity_check(Config(feature_type='r21d_rgb', extraction_fps=10.0))
+    with pytest.raises(ValueError):
+        sanity_check(Config(on_extraction='save_numpy', output_path='./x',
+                            tmp_path='./x'))
+    with pytest.raises(ValueError):
+        sanity_check(Config(show_pred=True, device_ids=[0, 1]))
+
+
+def test_cli_parser_matches_reference_flags():
+    from video_features_amd.cli import build_parser
+    p = build_parser()
+    args = p.parse_args(['--feature_type', 'CLIP-ViT-B/32',
+                         '--video_paths', 'a.mp4', 'b.mp4',
+                         '--device_ids', '0', '1',
+                         '--extract_method', 'uni_12',
+                         '--on_extraction', 'save_numpy',
+                         '--output_path', '/tmp/out',
+                         '--flow_type', 'raft',
+                         '--batch_size', '32',
+                         '--resize_to_larger_edge',
+                         '--side_size', '256'])
+    cfg = Config.coerce(args)
+    assert cfg.device_ids == [0, 1]
+    assert cfg.resize_to_smaller_edge is False
+    assert cfg.batch_size == 32
+
+
+def test_save_numpy_naming(tmp_path):
+    feats = {'clip': np.ones((4, 8), np.float32),
+             'fps': np.array(25.0),
+             'timestamps_ms': np.array([0.0, 40.0])}
+    out = str(tmp_path / 'out')
+    action_on_extraction(feats, '/data/videos/myvid.mp4', out, 'save_numpy',
+                         output_direct=False, feature_type='clip')
+    assert sorted(os.listdir(out)) == ['myvid_clip.npy']
+    arr = np.load(os.path.join(out, 'myvid_clip.npy'))
+    assert arr.shape == (4, 8)
+    assert outputs_exist(feats.keys(), '/data/videos/myvid.mp4', out,
+                         'save_numpy', False, 'clip')
+
+
+def test_save_numpy_direct_naming(tmp_path):
+    feats = {'clip': np.zeros((2, 2))}
+    out = str(tmp_path)
+    action_on_extraction(feats, 'v.mp4', out, 'save_numpy',
+                         output_direct=True, feature_type='clip')
+    assert os.path.exists(os.path.join(out, 'v.npy'))
+
+
+def test_save_pickle(tmp_path):
+    feats = {'i3d_rgb': np.ones((3, 1024)), 'fps': np.array(25.0)}
+    out = str(tmp_path)
+    action_on_extraction(feats, 'vid.avi', out, 'save_pickle',
+                         output_direct=False, feature_type='i3d')
+    with open(os.path.join(out, 'vid_i3d.pkl'), 'rb') as f:
+        loaded = pickle.load(f)
+    assert loaded['i3d_rgb'].shape == (3, 1024)
+
+
+def test_make_output_path():
+    assert make_output_path('/o', 'clip', False) == '/o/clip'
+    assert make_output_path('/o', 'clip', True) == '/o'
+
+
+def test_print_sink_smoke(capsys):
+    action_on_extraction({'f': np.arange(6.0).reshape(2, 3)}, 'v.mp4', '.',
+                         'print', False, 'f')
+    out = capsys.readouterr().out
+    assert 'shape (2, 3)' in out

@@ -1,0 +1,120 @@
+"""BaseExtractor — the shared pipeline controller every feature family uses.
+
+The reference has *no* shared engine: each of its seven extractors
+re-implements the decode→transform→infer→sink loop with copy-pasted
+variations (see reference models/CLIP/extract_clip.py:22-135 and its six
+siblings).  Here the loop lives once, and each family implements only
+``build_models(device, dtype)`` and ``extract(device, path)``.
+
+Preserved reference contracts:
+- constructor ``ExtractX(args, external_call=False)`` where ``args`` is any
+  namespace-like object (reference README.md:39-51);
+- ``forward(indices: LongTensor)`` — indices into the resolved path list,
+  ``indices.device`` selects the GPU (reference extract_clip.py:38-66);
+- per-video try/except that logs and continues, KeyboardInterrupt re-raised
+  (reference extract_clip.py:70-84);
+- ``external_call=True`` returns the feats list instead of sinking;
+- every feats dict carries ``fps`` and ``timestamps_ms`` meta keys.
+"""
+from __future__ import annotations
+
+import traceback
+from typing import Any, Dict, List, Optional
+
+import numpy as np
+import torch
+
+from ..config import Config
+from ..io.listing import form_list_from_user_input
+from ..runtime.sinks import (action_on_extraction, make_output_path,
+                             outputs_exist)
+from ..runtime.progress import make_progress
+
+
+class BaseExtractor(torch.nn.Module):
+    feature_type: str = ''
+
+    def __init__(self, args: Any, external_call: bool = False):
+        super().__init__()
+        cfg = Config.coerce(args)
+        self.cfg = cfg
+        self.feature_type = cfg.feature_type
+        self.path_list = form_list_from_user_input(cfg)
+        self.extraction_fps = cfg.extraction_fps
+        self.extract_method = cfg.extract_method
+        self.on_extraction = cfg.on_extraction
+        self.external_call = external_call
+        self.output_direct = cfg.output_direct
+        self.output_path = make_output_path(cfg.output_path, self.feature_type,
+                                            cfg.output_direct)
+        self.tmp_path = cfg.tmp_path
+        self.keep_tmp_files = cfg.keep_tmp_files
+        self.show_pred = cfg.show_pred
+        self.progress = make_progress(total=len(self.path_list))
+        self._models_cache: Dict[str, Any] = {}
+
+    # -------------------------------------------------------------- hooks
+    def build_models(self, device: torch.device, dtype: torch.dtype) -> Any:
+        """Construct (and move to device) everything ``extract`` needs."""
+        raise NotImplementedError
+
+    def extract(self, device: torch.device, models: Any,
+                video_path) -> Dict[str, np.ndarray]:
+        raise NotImplementedError
+
+    # ------------------------------------------------------------ helpers
+    def compute_dtype(self, device: torch.device) -> torch.dtype:
+        d = self.cfg.dtype
+        if d == 'fp32':
+            return torch.float32
+        if d == 'bf16':
+            return torch.bfloat16
+        return torch.bfloat16 if device.type == 'cuda' else torch.float32
+
+    def models_for(self, device: torch.device) -> Any:
+        key = str(device)
+        if key not in self._models_cache:
+            dtype = self.compute_dtype(device)
+            torch.manual_seed(self.cfg.seed)
+            self._models_cache[key] = self.build_models(device, dtype)
+        return self._models_cache[key]
+
+    # ------------------------------------------------------------ forward
+    @torch.no_grad()
+    def forward(self, indices: torch.LongTensor) -> List[Dict[str, np.ndarray]]:
+        device = indices.device
+        models = self.models_for(device)
+        feats_list: List[Dict[str, np.ndarray]] = []
+        for idx in indices.tolist():
+            video_path = self.path_list[idx]
+            try:
+                if (self.cfg.resume and not self.external_call
+                        and self._already_done(video_path)):
+                    self.progress.update()
+                    continue
+                feats_dict = self.extract(device, models, video_path)
+                if self.external_call:
+                    feats_list.append(feats_dict)
+                else:
+                    action_on_extraction(feats_dict, self._stem_path(video_path),
+                                         self.output_path, self.on_extraction,
+                                         self.output_direct, self.feature_type)
+            except KeyboardInterrupt:
+                raise
+            except Exception as e:
+                print(e)
+                print(f'Extraction failed at: {video_path} with error (^). '
+                      'Continuing extraction')
+                traceback.print_exc()
+            self.progress.update()
+        return feats_list
+
+    def _stem_path(self, video_path) -> str:
+        # pre-computed-flow inputs are (video, flow_dir) tuples
+        return video_path[0] if isinstance(video_path, tuple) else video_path
+
+    def _already_done(self, video_path) -> bool:
+        keys = getattr(self, 'output_feat_keys', [self.feature_type])
+        return outputs_exist(keys, self._stem_path(video_path), self.output_path,
+                             self.on_extraction, self.output_direct,
+                             self.feature_type)

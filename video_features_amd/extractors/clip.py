@@ -1,0 +1,56 @@
+"""CLIP feature extractor: 512-d embedding per sampled frame.
+
+Capability parity with the reference's ``ExtractCLIP``
+(reference models/CLIP/extract_clip.py): ``uni_N``/``fix_N`` frame sampling,
+feature shape (N, 512), meta keys fps/timestamps_ms, ``external_call`` API,
+feature types CLIP-ViT-B/32, CLIP-ViT-B/16, CLIP4CLIP-ViT-B-32 (the
+CLIP4CLIP variant loads task-tuned ViT-B/32 weights from ``weights_path``).
+"""
+from __future__ import annotations
+
+from typing import Dict
+
+import numpy as np
+import torch
+
+from .. import transforms as T
+from ..io.sampling import sample_indices, timestamps_ms
+from ..io.video import open_video
+from ..models.clip_vit import build_clip_vit
+from .base import BaseExtractor
+
+
+class ExtractCLIP(BaseExtractor):
+    def __init__(self, args, external_call: bool = False):
+        super().__init__(args, external_call)
+        self.extract_method = self.cfg.extract_method or 'uni_12'
+        self.batch_size = max(1, self.cfg.batch_size)
+
+    def build_models(self, device: torch.device, dtype: torch.dtype):
+        model = build_clip_vit(self.feature_type)
+        if self.cfg.weights_path:
+            sd = torch.load(self.cfg.weights_path, map_location='cpu',
+                            weights_only=True)
+            model.load_state_dict(sd)
+        model = model.to(device=device, dtype=dtype).eval()
+        return model
+
+    def extract(self, device: torch.device, model,
+                video_path) -> Dict[str, np.ndarray]:
+        reader = open_video(video_path, self.tmp_path, self.extraction_fps)
+        fps = reader.fps
+        idxs = sample_indices(self.extract_method, reader.frame_count, fps)
+        frames_u8 = torch.from_numpy(reader.read_frames(idxs))
+        batch = T.clip_preprocess(frames_u8, model.cfg.input_resolution)
+        dtype = self.compute_dtype(device)
+        feats = []
+        for s in range(0, batch.shape[0], max(self.batch_size, 16)):
+            chunk = batch[s:s + max(self.batch_size, 16)]
+            chunk = chunk.to(device=device, dtype=dtype, non_blocking=True)
+            feats.append(model.encode_image(chunk).float().cpu())
+        features = torch.cat(feats).numpy()
+        return {
+            self.feature_type: features,
+            'fps': np.array(fps),
+            'timestamps_ms': np.array(timestamps_ms(idxs, fps)),
+        }

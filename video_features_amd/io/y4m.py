@@ -1,0 +1,133 @@
+"""Minimal YUV4MPEG2 (.y4m) reader/writer — a zero-dependency raw video format.
+
+Lets the framework own a decode path with no ffmpeg/mmcv/OpenCV in the image
+(the reference shells out to ffmpeg and uses mmcv.VideoReader — reference
+utils/utils.py:207-276, 310).  Y4M is uncompressed 4:2:0 YUV with a text
+header, trivially parseable, and ffmpeg can produce it anywhere, so it doubles
+as the interchange format between an external ffmpeg and this framework.
+"""
+from __future__ import annotations
+
+import io
+import os
+from typing import List, Tuple
+
+import numpy as np
+
+_MAGIC = b'YUV4MPEG2'
+
+
+def _yuv420_to_rgb(y: np.ndarray, u: np.ndarray, v: np.ndarray) -> np.ndarray:
+    """BT.601 full-swing-ish conversion used by the JPEG/JFIF convention."""
+    h, w = y.shape
+    u_full = np.repeat(np.repeat(u, 2, axis=0), 2, axis=1)[:h, :w]
+    v_full = np.repeat(np.repeat(v, 2, axis=0), 2, axis=1)[:h, :w]
+    yf = y.astype(np.float32)
+    uf = u_full.astype(np.float32) - 128.0
+    vf = v_full.astype(np.float32) - 128.0
+    r = yf + 1.402 * vf
+    g = yf - 0.344136 * uf - 0.714136 * vf
+    b = yf + 1.772 * uf
+    return np.clip(np.stack([r, g, b], axis=-1), 0, 255).astype(np.uint8)
+
+
+def _rgb_to_yuv420(rgb: np.ndarray) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
+    f = rgb.astype(np.float32)
+    r, g, b = f[..., 0], f[..., 1], f[..., 2]
+    y = 0.299 * r + 0.587 * g + 0.114 * b
+    u = -0.168736 * r - 0.331264 * g + 0.5 * b + 128.0
+    v = 0.5 * r - 0.418688 * g - 0.081312 * b + 128.0
+    y8 = np.clip(y, 0, 255).astype(np.uint8)
+    # average 2x2 blocks for chroma subsampling (pad odd dims)
+    h, w = y8.shape
+    hp, wp = h + (h & 1), w + (w & 1)
+    up = np.zeros((hp, wp), np.float32)
+    vp = np.zeros((hp, wp), np.float32)
+    up[:h, :w], vp[:h, :w] = u, v
+    if h & 1:
+        up[h, :w], vp[h, :w] = u[h - 1], v[h - 1]
+    if w & 1:
+        up[:h, w], vp[:h, w] = u[:, w - 1], v[:, w - 1]
+    u4 = up.reshape(hp // 2, 2, wp // 2, 2).mean(axis=(1, 3))
+    v4 = vp.reshape(hp // 2, 2, wp // 2, 2).mean(axis=(1, 3))
+    return y8, np.clip(u4, 0, 255).astype(np.uint8), np.clip(v4, 0, 255).astype(np.uint8)
+
+
+class Y4MReader:
+    """Index-addressable reader over a .y4m file (frames memory-mapped lazily)."""
+
+    def __init__(self, path: str):
+        self.path = path
+        with open(path, 'rb') as f:
+            header = f.readline()
+        if not header.startswith(_MAGIC):
+            raise ValueError(f'{path}: not a YUV4MPEG2 file')
+        self.width = self.height = 0
+        fps_num, fps_den = 25, 1
+        colorspace = 'C420'
+        for tok in header.split()[1:]:
+            t = tok.decode('ascii', 'replace')
+            if t.startswith('W'):
+                self.width = int(t[1:])
+            elif t.startswith('H'):
+                self.height = int(t[1:])
+            elif t.startswith('F'):
+                num, den = t[1:].split(':')
+                fps_num, fps_den = int(num), int(den)
+            elif t.startswith('C'):
+                colorspace = t
+        if not colorspace.startswith('C420'):
+            raise ValueError(f'{path}: only 4:2:0 y4m supported, got {colorspace}')
+        self.fps = fps_num / fps_den
+        self._header_len = len(header)
+        y_sz = self.width * self.height
+        c_sz = ((self.width + 1) // 2) * ((self.height + 1) // 2)
+        self._frame_data = y_sz + 2 * c_sz
+        self._y_sz, self._c_sz = y_sz, c_sz
+        # frame record = b'FRAME...\n' + planes; assume constant FRAME header len
+        with open(path, 'rb') as f:
+            f.seek(self._header_len)
+            fh = f.readline()
+        if not fh.startswith(b'FRAME'):
+            raise ValueError(f'{path}: malformed y4m (no FRAME marker)')
+        self._frame_hdr_len = len(fh)
+        total = os.path.getsize(path) - self._header_len
+        rec = self._frame_hdr_len + self._frame_data
+        self.frame_count = total // rec
+        self._rec = rec
+
+    def read_frame(self, idx: int) -> np.ndarray:
+        if not (0 <= idx < self.frame_count):
+            raise IndexError(f'frame {idx} out of range [0, {self.frame_count})')
+        off = self._header_len + idx * self._rec + self._frame_hdr_len
+        with open(self.path, 'rb') as f:
+            f.seek(off)
+            raw = f.read(self._frame_data)
+        w, h = self.width, self.height
+        cw, ch = (w + 1) // 2, (h + 1) // 2
+        y = np.frombuffer(raw, np.uint8, self._y_sz).reshape(h, w)
+        u = np.frombuffer(raw, np.uint8, self._c_sz, offset=self._y_sz).reshape(ch, cw)
+        v = np.frombuffer(raw, np.uint8, self._c_sz, offset=self._y_sz + self._c_sz).reshape(ch, cw)
+        return _yuv420_to_rgb(y, u, v)
+
+    def read_frames(self, indices) -> np.ndarray:
+        return np.stack([self.read_frame(int(i)) for i in indices])
+
+
+def write_y4m(path: str, frames: np.ndarray, fps: float = 25.0) -> None:
+    """Write (T, H, W, 3) uint8 RGB frames as a 4:2:0 y4m file."""
+    frames = np.asarray(frames)
+    if frames.ndim != 4 or frames.shape[-1] != 3:
+        raise ValueError(f'expected (T,H,W,3), got {frames.shape}')
+    t, h, w, _ = frames.shape
+    from fractions import Fraction
+    fr = Fraction(fps).limit_denominator(1001)
+    with open(path, 'wb') as f:
+        f.write(f'YUV4MPEG2 W{w} H{h} F{fr.numerator}:{fr.denominator} Ip A1:1 C420jpeg\n'
+                .encode('ascii'))
+        for i in range(t):
+            y, u, v = _rgb_to_yuv420(frames[i])
+            f.write(b'FRAME\n')
+            f.write(y.tobytes())
+            f.write(u.tobytes())
+            f.write(v.tobytes())
