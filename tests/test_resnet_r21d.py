@@ -1,0 +1,66 @@
+import numpy as np
+import pytest
+import torch
+
+from video_features_amd.config import Config
+
+
+def test_resnet_shapes():
+    from video_features_amd.models.resnet import build_resnet
+    torch.manual_seed(0)
+    for name, dim in [('resnet18', 512), ('resnet50', 2048)]:
+        m = build_resnet(name).eval()
+        with torch.no_grad():
+            f = m.forward_features(torch.randn(2, 3, 224, 224))
+            logits = m(torch.randn(1, 3, 224, 224))
+        assert f.shape == (2, dim)
+        assert logits.shape == (1, 1000)
+
+
+def test_resnet_param_counts():
+    # sanity vs the published architectures (±1% of torchvision counts)
+    from video_features_amd.models.resnet import build_resnet
+    expected = {'resnet18': 11.69e6, 'resnet34': 21.80e6, 'resnet50': 25.56e6,
+                'resnet101': 44.55e6, 'resnet152': 60.19e6}
+    for name, exp in expected.items():
+        n = sum(p.numel() for p in build_resnet(name).parameters())
+        assert abs(n - exp) / exp < 0.01, (name, n, exp)
+
+
+def test_r21d_shapes_and_params():
+    from video_features_amd.models.r21d import R2Plus1D18
+    torch.manual_seed(0)
+    m = R2Plus1D18().eval()
+    n = sum(p.numel() for p in m.parameters())
+    # torchvision r2plus1d_18 has 31.51M params
+    assert abs(n - 31.51e6) / 31.51e6 < 0.01, n
+    with torch.no_grad():
+        f = m.forward_features(torch.randn(1, 3, 16, 112, 112))
+    assert f.shape == (1, 512)
+
+
+def test_resnet_extractor_end_to_end(y4m_video):
+    from video_features_amd.extractors.resnet import ExtractResNet
+    cfg = Config(feature_type='resnet18', video_paths=[y4m_video],
+                 batch_size=8, cpu=True)
+    ex = ExtractResNet(cfg, external_call=True)
+    out = ex(torch.arange(1))[0]
+    assert out['resnet18'].shape == (16, 512)
+    assert len(out['timestamps_ms']) == 16
+    assert np.isfinite(out['resnet18']).all()
+
+
+def test_r21d_extractor_end_to_end(y4m_video):
+    from video_features_amd.extractors.r21d import ExtractR21D
+    cfg = Config(feature_type='r21d_rgb', video_paths=[y4m_video], cpu=True)
+    ex = ExtractR21D(cfg, external_call=True)
+    out = ex(torch.arange(1))[0]
+    # 16 frames, stack 16 step 16 → exactly 1 window
+    assert out['r21d_rgb'].shape == (1, 512)
+    assert np.isfinite(out['r21d_rgb']).all()
+
+
+def test_show_pred_labels():
+    from video_features_amd.utils.labels import class_names
+    assert len(class_names('kinetics')) == 400
+    assert len(class_names('imagenet')) == 1000

@@ -1,0 +1,59 @@
+"""ResNet frame-wise features: (num_frames, 512/2048) per video.
+
+Capability parity with the reference's ``ExtractResNet``
+(reference models/resnet/extract_resnet.py): streaming decode with
+batch-size accumulation, ImageNet eval preprocessing, per-frame
+``timestamps_ms``, ``--show_pred`` top-5 over ImageNet.
+"""
+from __future__ import annotations
+
+from typing import Dict
+
+import numpy as np
+import torch
+
+from .. import transforms as T
+from ..io.sampling import timestamps_ms
+from ..io.video import open_video
+from ..models.resnet import build_resnet
+from .base import BaseExtractor
+
+
+class ExtractResNet(BaseExtractor):
+    def __init__(self, args, external_call: bool = False):
+        super().__init__(args, external_call)
+        self.batch_size = max(1, self.cfg.batch_size)
+
+    def build_models(self, device: torch.device, dtype: torch.dtype):
+        model = build_resnet(self.feature_type)
+        if self.cfg.weights_path:
+            sd = torch.load(self.cfg.weights_path, map_location='cpu',
+                            weights_only=True)
+            model.load_state_dict(sd)
+        return model.to(device=device, dtype=dtype).eval()
+
+    def extract(self, device: torch.device, model,
+                video_path) -> Dict[str, np.ndarray]:
+        reader = open_video(video_path, self.tmp_path, self.extraction_fps)
+        fps = reader.fps
+        dtype = self.compute_dtype(device)
+        feats, idx_done = [], []
+        n = reader.frame_count
+        for start in range(0, n, self.batch_size):
+            idxs = list(range(start, min(start + self.batch_size, n)))
+            frames_u8 = torch.from_numpy(reader.read_frames(idxs))
+            batch = T.imagenet_preprocess(frames_u8)
+            batch = batch.to(device=device, dtype=dtype, non_blocking=True)
+            feats.append(model.forward_features(batch).float().cpu())
+            idx_done.extend(idxs)
+            if self.show_pred:
+                from ..utils.labels import show_predictions_on_dataset
+                show_predictions_on_dataset(model.fc(feats[-1].to(device=device,
+                                                                  dtype=dtype)),
+                                            'imagenet')
+        features = torch.cat(feats).numpy()
+        return {
+            self.feature_type: features,
+            'fps': np.array(fps),
+            'timestamps_ms': np.array(timestamps_ms(np.array(idx_done), fps)),
+        }

@@ -1,0 +1,112 @@
+"""Native ResNet-18/34/50/101/152 (feature extractor + classifier head).
+
+The reference pulls these from the torchvision model zoo (reference
+models/resnet/extract_resnet.py:54-67).  From-scratch implementation of the
+standard v1 architecture; ``forward_features`` returns the post-avgpool
+embedding (512-d for 18/34, 2048-d for 50/101/152) and ``forward`` the
+1000-way ImageNet logits used by ``--show_pred``.
+
+Convolutions run through PyTorch-ROCm (MIOpen); on MI355X the surrounding
+normalization/activation work is fused by channels_last memory format.
+"""
+from __future__ import annotations
+
+from typing import List, Type
+
+import torch
+from torch import nn
+
+
+class BasicBlock(nn.Module):
+    expansion = 1
+
+    def __init__(self, in_planes: int, planes: int, stride: int = 1):
+        super().__init__()
+        self.conv1 = nn.Conv2d(in_planes, planes, 3, stride, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, 1, 1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = None
+        if stride != 1 or in_planes != planes:
+            self.downsample = nn.Sequential(
+                nn.Conv2d(in_planes, planes, 1, stride, bias=False),
+                nn.BatchNorm2d(planes))
+
+    def forward(self, x):
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.bn2(self.conv2(out))
+        return self.relu(out + identity)
+
+
+class Bottleneck(nn.Module):
+    expansion = 4
+
+    def __init__(self, in_planes: int, planes: int, stride: int = 1):
+        super().__init__()
+        out_planes = planes * self.expansion
+        self.conv1 = nn.Conv2d(in_planes, planes, 1, bias=False)
+        self.bn1 = nn.BatchNorm2d(planes)
+        self.conv2 = nn.Conv2d(planes, planes, 3, stride, 1, bias=False)
+        self.bn2 = nn.BatchNorm2d(planes)
+        self.conv3 = nn.Conv2d(planes, out_planes, 1, bias=False)
+        self.bn3 = nn.BatchNorm2d(out_planes)
+        self.relu = nn.ReLU(inplace=True)
+        self.downsample = None
+        if stride != 1 or in_planes != out_planes:
+            self.downsample = nn.Sequential(
+                nn.Conv2d(in_planes, out_planes, 1, stride, bias=False),
+                nn.BatchNorm2d(out_planes))
+
+    def forward(self, x):
+        identity = x if self.downsample is None else self.downsample(x)
+        out = self.relu(self.bn1(self.conv1(x)))
+        out = self.relu(self.bn2(self.conv2(out)))
+        out = self.bn3(self.conv3(out))
+        return self.relu(out + identity)
+
+
+class ResNet(nn.Module):
+    def __init__(self, block: Type, layers: List[int], num_classes: int = 1000):
+        super().__init__()
+        self.conv1 = nn.Conv2d(3, 64, 7, 2, 3, bias=False)
+        self.bn1 = nn.BatchNorm2d(64)
+        self.relu = nn.ReLU(inplace=True)
+        self.maxpool = nn.MaxPool2d(3, 2, 1)
+        self.in_planes = 64
+        self.layer1 = self._make_layer(block, 64, layers[0], 1)
+        self.layer2 = self._make_layer(block, 128, layers[1], 2)
+        self.layer3 = self._make_layer(block, 256, layers[2], 2)
+        self.layer4 = self._make_layer(block, 512, layers[3], 2)
+        self.avgpool = nn.AdaptiveAvgPool2d(1)
+        self.feat_dim = 512 * block.expansion
+        self.fc = nn.Linear(self.feat_dim, num_classes)
+
+    def _make_layer(self, block, planes, n, stride):
+        blocks = [block(self.in_planes, planes, stride)]
+        self.in_planes = planes * block.expansion
+        blocks += [block(self.in_planes, planes) for _ in range(n - 1)]
+        return nn.Sequential(*blocks)
+
+    def forward_features(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
+        return self.avgpool(x).flatten(1)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return self.fc(self.forward_features(x))
+
+
+_SPECS = {
+    'resnet18': (BasicBlock, [2, 2, 2, 2]),
+    'resnet34': (BasicBlock, [3, 4, 6, 3]),
+    'resnet50': (Bottleneck, [3, 4, 6, 3]),
+    'resnet101': (Bottleneck, [3, 4, 23, 3]),
+    'resnet152': (Bottleneck, [3, 8, 36, 3]),
+}
+
+
+def build_resnet(name: str, num_classes: int = 1000) -> ResNet:
+    block, layers = _SPECS[name]
+    return ResNet(block, layers, num_classes)
