@@ -1,0 +1,137 @@
+"""I3D two-stream extractor: per 64-frame stack, 1024-d per stream.
+
+Capability parity with the reference's ``ExtractI3D``
+(reference models/i3d/extract_i3d.py): rgb+flow streams (``--streams``),
+flow from RAFT / PWC / pre-computed flow JPG dirs (``--flow_type``),
+stack/step 64 sliding windows needing stack+1 frames (flow is between
+consecutive frames), short videos resampled to stack+1 frames by linspace,
+smaller-edge resize to 256 before flow, center-crop 224 + [-1, 1] scaling
+per stream, ``--show_pred`` Kinetics top-5.  Output keys: ``rgb`` /
+``flow`` (+ fps, timestamps_ms).
+"""
+from __future__ import annotations
+
+from pathlib import Path
+from typing import Dict, List, Optional
+
+import numpy as np
+import torch
+
+from .. import transforms as T
+from ..io.video import open_video, ImageDirReader
+from ..models.i3d import I3D
+from ..models.pwc import PWCNet
+from ..models.raft import RAFT, InputPadder
+from .base import BaseExtractor
+
+DEFAULT_STACK_SIZE = 64
+DEFAULT_STEP_SIZE = 64
+RESIZE_SIDE = 256
+CROP = 224
+
+
+class ExtractI3D(BaseExtractor):
+    output_feat_keys = ['rgb', 'flow']
+
+    def __init__(self, args, external_call: bool = False):
+        super().__init__(args, external_call)
+        self.streams = self.cfg.streams or ['rgb', 'flow']
+        self.stack_size = self.cfg.stack_size or DEFAULT_STACK_SIZE
+        self.step_size = self.cfg.step_size or DEFAULT_STEP_SIZE
+        self.flow_type = self.cfg.flow_type
+        self.output_feat_keys = list(self.streams)
+
+    def build_models(self, device: torch.device, dtype: torch.dtype):
+        models = {}
+        if 'rgb' in self.streams:
+            models['rgb'] = I3D(modality='rgb').to(device, dtype).eval()
+        if 'flow' in self.streams:
+            models['flow'] = I3D(modality='flow').to(device, dtype).eval()
+            if self.flow_type == 'raft':
+                models['flow_xtr'] = RAFT().to(device, dtype).eval()
+            elif self.flow_type == 'pwc':
+                models['flow_xtr'] = PWCNet().to(device, dtype).eval()
+        return models
+
+    # ------------------------------------------------------------ helpers
+    def _read_resized_frames(self, video_path) -> (torch.Tensor, float):
+        reader = open_video(video_path, self.tmp_path, self.extraction_fps)
+        n = reader.frame_count
+        need = self.stack_size + 1
+        if n < need:
+            # short video: resample to stack+1 frames by linspace
+            # (reference extract_i3d.py:250-255)
+            idxs = np.linspace(0, n - 1, need).round().astype(np.int64)
+        else:
+            idxs = np.arange(n)
+        frames_u8 = torch.from_numpy(reader.read_frames(idxs))
+        x = frames_u8.permute(0, 3, 1, 2).float()   # (T, 3, H, W), [0, 255]
+        x = T.resize_improved(x, RESIZE_SIDE, smaller_edge=True)
+        return x, reader.fps
+
+    def _compute_flow(self, models, stack: torch.Tensor) -> torch.Tensor:
+        """(S+1, 3, H, W) frames → (S, 2, H, W) flow via RAFT or PWC."""
+        if self.flow_type == 'raft':
+            padder = InputPadder(stack.shape)
+            im1, im2 = padder.pad(stack[:-1], stack[1:])
+            return padder.unpad(models['flow_xtr'](im1, im2, test_mode=True))
+        return models['flow_xtr'](stack[:-1], stack[1:])
+
+    def _read_precomputed_flow(self, flow_dir: str, device, dtype,
+                               count: int) -> torch.Tensor:
+        """flow_x_*.jpg / flow_y_*.jpg dirs → (T, 2, H, W) float flow in
+        [-20, 20] (reference extract_i3d.py:231-237, 266-278)."""
+        xs = ImageDirReader(flow_dir, pattern=r'flow_x').read_frames(range(count))
+        ys = ImageDirReader(flow_dir, pattern=r'flow_y').read_frames(range(count))
+        fx = torch.from_numpy(xs[..., 0]).float()
+        fy = torch.from_numpy(ys[..., 0]).float()
+        flow = torch.stack([fx, fy], dim=1)          # (T, 2, H, W) in [0,255]
+        flow = flow / 255.0 * 40.0 - 20.0
+        x = T.resize_improved(flow, RESIZE_SIDE, smaller_edge=True)
+        return x.to(device=device, dtype=dtype)
+
+    # ------------------------------------------------------------ extract
+    def extract(self, device: torch.device, models,
+                video_path) -> Dict[str, np.ndarray]:
+        precomputed = isinstance(video_path, tuple)
+        vid_path = video_path[0] if precomputed else video_path
+        frames, fps = self._read_resized_frames(vid_path)
+        dtype = self.compute_dtype(device)
+        n = frames.shape[0]
+        feats: Dict[str, List] = {s: [] for s in self.streams}
+        ts: List[float] = []
+        flow_all: Optional[torch.Tensor] = None
+        if precomputed and 'flow' in self.streams:
+            flow_all = self._read_precomputed_flow(video_path[1], device,
+                                                   dtype, n - 1)
+        start = 0
+        while start + self.stack_size + 1 <= n:
+            stack = frames[start:start + self.stack_size + 1]
+            stack = stack.to(device=device, dtype=dtype, non_blocking=True)
+            for stream in self.streams:
+                if stream == 'rgb':
+                    x = T.center_crop(stack[:-1], CROP)
+                    x = T.scale_to_pm1(x)
+                else:
+                    if flow_all is not None:
+                        flow = flow_all[start:start + self.stack_size]
+                    else:
+                        flow = self._compute_flow(models, stack)
+                    x = T.i3d_flow_preprocess(flow, CROP)
+                clip = x.transpose(0, 1)[None]       # (1, C, T, H, W)
+                f = models[stream].forward_features(clip)
+                feats[stream].append(f.float().cpu())
+                if self.show_pred:
+                    from ..utils.labels import show_predictions_on_dataset
+                    print(f'{stream} stack @ {start}:')
+                    show_predictions_on_dataset(models[stream](clip).float().cpu(),
+                                                'kinetics')
+            ts.append(start / fps * 1000.0)
+            start += self.step_size
+        out: Dict[str, np.ndarray] = {
+            s: (torch.cat(feats[s]).numpy() if feats[s]
+                else np.zeros((0, I3D.FEAT_DIM), np.float32))
+            for s in self.streams}
+        out['fps'] = np.array(fps)
+        out['timestamps_ms'] = np.array(ts)
+        return out

@@ -1,0 +1,52 @@
+"""VGGish audio extractor: (Ta, 128), one row per 0.96 s of audio.
+
+Capability parity with BOTH reference variants — the TF1 ``vggish``
+(reference models/vggish/extract_vggish.py) and torch ``vggish_torch``
+(reference models/vggish_torch/extract_vggish.py) — served by the single
+native implementation in models/vggish.py.  Audio arrives from a .wav
+input, a sidecar .wav, or ffmpeg extraction (mp4→aac→wav, reference
+utils/utils.py:247-276); tmp files are removed unless ``--keep_tmp_files``.
+"""
+from __future__ import annotations
+
+import os
+from typing import Dict
+
+import numpy as np
+import torch
+
+from ..io.audio import load_audio_for_video, resample
+from ..models.vggish import SAMPLE_RATE, VGGish, waveform_to_examples
+from .base import BaseExtractor
+
+
+class ExtractVGGish(BaseExtractor):
+    def build_models(self, device: torch.device, dtype: torch.dtype):
+        model = VGGish(postprocess=False)
+        if self.cfg.weights_path:
+            sd = torch.load(self.cfg.weights_path, map_location='cpu',
+                            weights_only=True)
+            model.load_state_dict(sd)
+        return model.to(device=device, dtype=dtype).eval()
+
+    def extract(self, device: torch.device, model,
+                video_path) -> Dict[str, np.ndarray]:
+        samples, sr, tmp_files = load_audio_for_video(video_path, self.tmp_path,
+                                                      self.keep_tmp_files)
+        try:
+            samples = resample(samples, sr, SAMPLE_RATE)
+            wav = torch.from_numpy(np.ascontiguousarray(samples))
+            examples = waveform_to_examples(wav, device)
+            dtype = self.compute_dtype(device)
+            feats = model(examples.to(dtype)).float().cpu().numpy()
+        finally:
+            if not self.keep_tmp_files:
+                for f in tmp_files:
+                    if os.path.exists(f):
+                        os.remove(f)
+        n = feats.shape[0]
+        return {
+            self.feature_type: feats,
+            'fps': np.array(float(SAMPLE_RATE)),
+            'timestamps_ms': np.array([i * 960.0 for i in range(n)]),
+        }

@@ -1,0 +1,147 @@
+"""Native I3D (Inflated Inception-3D) for RGB and flow streams.
+
+Re-implementation of the architecture the reference vendors
+(reference models/i3d/i3d_src/i3d_net.py): Inception-v1 inflated to 3D with
+TensorFlow-SAME padding semantics (asymmetric pads computed from the input
+size), nine Mixed blocks, and a features mode that returns the 1024-d
+embedding time-averaged over the clip
+(reference i3d_net.py:238-264).
+
+TF-SAME: for kernel k, stride s, input n the total pad is
+``max(k - s, 0)`` when ``n % s == 0`` else ``max(k - n % s, 0)``, split with
+the extra cell at the *end* — this differs from PyTorch's symmetric padding
+and changes borders, so it is computed dynamically in forward via F.pad.
+"""
+from __future__ import annotations
+
+from typing import List, Tuple
+
+import torch
+import torch.nn.functional as F
+from torch import nn
+
+
+def _same_pad_1d(n: int, k: int, s: int) -> Tuple[int, int]:
+    total = max(k - s, 0) if n % s == 0 else max(k - (n % s), 0)
+    front = total // 2
+    return front, total - front
+
+
+def tf_same_pad_3d(x: torch.Tensor, kernel, stride) -> torch.Tensor:
+    """Pad (B, C, T, H, W) with TF-SAME semantics for a 3D conv/pool."""
+    t, h, w = x.shape[-3:]
+    pt = _same_pad_1d(t, kernel[0], stride[0])
+    ph = _same_pad_1d(h, kernel[1], stride[1])
+    pw = _same_pad_1d(w, kernel[2], stride[2])
+    # F.pad order: (w_lo, w_hi, h_lo, h_hi, t_lo, t_hi)
+    return F.pad(x, (pw[0], pw[1], ph[0], ph[1], pt[0], pt[1]))
+
+
+def _triple(v) -> Tuple[int, int, int]:
+    return (v, v, v) if isinstance(v, int) else tuple(v)
+
+
+class Unit3D(nn.Module):
+    """Conv3d + BN + ReLU with TF-SAME padding
+    (the reference's ``Unit3Dpy``, i3d_net.py:37-105)."""
+
+    def __init__(self, in_ch: int, out_ch: int, kernel=1, stride=1,
+                 use_bn: bool = True, activation: bool = True,
+                 use_bias: bool = False):
+        super().__init__()
+        self.kernel = _triple(kernel)
+        self.stride = _triple(stride)
+        self.conv = nn.Conv3d(in_ch, out_ch, self.kernel, self.stride,
+                              padding=0, bias=use_bias)
+        self.bn = nn.BatchNorm3d(out_ch, eps=1e-3) if use_bn else None
+        self.activation = activation
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.conv(tf_same_pad_3d(x, self.kernel, self.stride))
+        if self.bn is not None:
+            x = self.bn(x)
+        return F.relu(x, inplace=True) if self.activation else x
+
+
+class MaxPool3dSame(nn.Module):
+    def __init__(self, kernel, stride):
+        super().__init__()
+        self.kernel = _triple(kernel)
+        self.stride = _triple(stride)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return F.max_pool3d(tf_same_pad_3d(x, self.kernel, self.stride),
+                            self.kernel, self.stride)
+
+
+class Mixed(nn.Module):
+    """Inception block: 1×1 | 1×1→3×3 | 1×1→3×3 | pool→1×1
+    (reference i3d_net.py:123-157)."""
+
+    def __init__(self, in_ch: int, out: List[int]):
+        super().__init__()
+        self.b0 = Unit3D(in_ch, out[0], 1)
+        self.b1 = nn.Sequential(Unit3D(in_ch, out[1], 1),
+                                Unit3D(out[1], out[2], 3))
+        self.b2 = nn.Sequential(Unit3D(in_ch, out[3], 1),
+                                Unit3D(out[3], out[4], 3))
+        self.b3 = nn.Sequential(MaxPool3dSame(3, 1),
+                                Unit3D(in_ch, out[5], 1))
+        self.out_channels = out[0] + out[2] + out[4] + out[5]
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        return torch.cat([self.b0(x), self.b1(x), self.b2(x), self.b3(x)], 1)
+
+
+class I3D(nn.Module):
+    FEAT_DIM = 1024
+
+    def __init__(self, num_classes: int = 400, modality: str = 'rgb',
+                 dropout_prob: float = 0.0):
+        super().__init__()
+        in_ch = {'rgb': 3, 'flow': 2}[modality]
+        self.modality = modality
+        self.conv3d_1a_7x7 = Unit3D(in_ch, 64, 7, 2)
+        self.maxPool3d_2a_3x3 = MaxPool3dSame((1, 3, 3), (1, 2, 2))
+        self.conv3d_2b_1x1 = Unit3D(64, 64, 1)
+        self.conv3d_2c_3x3 = Unit3D(64, 192, 3)
+        self.maxPool3d_3a_3x3 = MaxPool3dSame((1, 3, 3), (1, 2, 2))
+        self.mixed_3b = Mixed(192, [64, 96, 128, 16, 32, 32])      # → 256
+        self.mixed_3c = Mixed(256, [128, 128, 192, 32, 96, 64])    # → 480
+        self.maxPool3d_4a_3x3 = MaxPool3dSame(3, 2)
+        self.mixed_4b = Mixed(480, [192, 96, 208, 16, 48, 64])     # → 512
+        self.mixed_4c = Mixed(512, [160, 112, 224, 24, 64, 64])    # → 512
+        self.mixed_4d = Mixed(512, [128, 128, 256, 24, 64, 64])    # → 512
+        self.mixed_4e = Mixed(512, [112, 144, 288, 32, 64, 64])    # → 528
+        self.mixed_4f = Mixed(528, [256, 160, 320, 32, 128, 128])  # → 832
+        self.maxPool3d_5a_2x2 = MaxPool3dSame(2, 2)
+        self.mixed_5b = Mixed(832, [256, 160, 320, 32, 128, 128])  # → 832
+        self.mixed_5c = Mixed(832, [384, 192, 384, 48, 128, 128])  # → 1024
+        self.dropout = nn.Dropout(dropout_prob)
+        self.conv3d_0c_1x1 = Unit3D(1024, num_classes, 1, use_bn=False,
+                                    activation=False, use_bias=True)
+
+    def _backbone(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.conv3d_1a_7x7(x)
+        x = self.maxPool3d_2a_3x3(x)
+        x = self.conv3d_2c_3x3(self.conv3d_2b_1x1(x))
+        x = self.maxPool3d_3a_3x3(x)
+        x = self.mixed_3c(self.mixed_3b(x))
+        x = self.maxPool3d_4a_3x3(x)
+        x = self.mixed_4f(self.mixed_4e(self.mixed_4d(self.mixed_4c(self.mixed_4b(x)))))
+        x = self.maxPool3d_5a_2x2(x)
+        x = self.mixed_5c(self.mixed_5b(x))
+        # avg pool (2, 7, 7), stride 1 (reference i3d_net.py:229-235)
+        kt = min(2, x.shape[2])
+        return F.avg_pool3d(x, (kt, min(7, x.shape[3]), min(7, x.shape[4])))
+
+    def forward_features(self, x: torch.Tensor) -> torch.Tensor:
+        """(B, C, T, H, W) → (B, 1024): spatial squeeze + mean over remaining
+        time (reference i3d_net.py:238-264, features=True)."""
+        x = self._backbone(x)
+        return x.mean(dim=(2, 3, 4))
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        x = self.dropout(self._backbone(x))
+        logits = self.conv3d_0c_1x1(x)          # (B, K, t, 1, 1)
+        return logits.mean(dim=(2, 3, 4))       # time-averaged class scores
