@@ -1,0 +1,70 @@
+"""Multi-process runtime tests over gloo (world_size 2, CPU) — the GPU-free
+coverage of the RCCL/xGMI path (same code, backend 'nccl' on MI355X)."""
+import os
+
+import numpy as np
+import pytest
+import torch
+
+from video_features_amd.config import Config
+from video_features_amd.runtime.dist import (resolve_devices, run_extraction,
+                                             shard_indices)
+
+
+def _make_videos(tmp_path, n=4):
+    from tests.conftest import synthetic_frames
+    from video_features_amd.io.y4m import write_y4m
+    paths = []
+    for i in range(n):
+        p = str(tmp_path / f'v{i}.y4m')
+        write_y4m(p, synthetic_frames(t=8, h=48, w=48, seed=i), fps=25.0)
+        paths.append(p)
+    return paths
+
+
+def test_shard_indices_cover_all():
+    a = shard_indices(10, 0, 3).tolist()
+    b = shard_indices(10, 1, 3).tolist()
+    c = shard_indices(10, 2, 3).tolist()
+    assert sorted(a + b + c) == list(range(10))
+    assert not (set(a) & set(b))
+
+
+def test_resolve_devices_cpu_multi():
+    cfg = Config(cpu=True, device_ids=[0, 1])
+    assert resolve_devices(cfg) == ['cpu', 'cpu']
+    assert resolve_devices(Config(cpu=True)) == ['cpu']
+
+
+@pytest.mark.timeout(600)
+def test_two_process_gloo_extraction(tmp_path):
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    paths = _make_videos(tmp_path, n=4)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=paths,
+                 extract_method='uni_2', cpu=True, device_ids=[0, 1],
+                 on_extraction='save_numpy',
+                 output_path=str(tmp_path / 'out'),
+                 tmp_path=str(tmp_path / 'tmp'))
+    run_extraction(cfg)
+    out_dir = tmp_path / 'out' / 'CLIP-ViT-B/32'
+    files = sorted(os.listdir(out_dir))
+    assert len(files) == 4, files   # every shard wrote its videos
+
+
+@pytest.mark.timeout(600)
+def test_two_process_gather_features(tmp_path):
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    paths = _make_videos(tmp_path, n=3)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=paths,
+                 extract_method='uni_2', cpu=True, device_ids=[0, 1],
+                 gather_features=True, tmp_path=str(tmp_path / 'tmp'))
+    feats = run_extraction(cfg)
+    assert len(feats) == 3
+    for fd in feats:
+        assert fd['CLIP-ViT-B/32'].shape == (2, 512)
+    # weight broadcast ⇒ both ranks computed with identical weights ⇒ the
+    # same video extracted single-process gives identical features
+    solo = run_extraction(cfg.replace(device_ids=None, gather_features=True))
+    for a, b in zip(feats, solo):
+        np.testing.assert_allclose(a['CLIP-ViT-B/32'], b['CLIP-ViT-B/32'],
+                                   rtol=1e-4, atol=1e-5)

@@ -40,6 +40,10 @@ def find_free_port() -> int:
 
 def resolve_devices(cfg: Config) -> List[str]:
     if cfg.cpu:
+        # --cpu with --device_ids N... runs that many CPU worker processes
+        # over gloo — the GPU-free test/CI path for the distributed runtime
+        if cfg.device_ids and len(cfg.device_ids) > 1:
+            return ['cpu'] * len(cfg.device_ids)
         return ['cpu']
     if cfg.device_ids:
         return [f'cuda:{i}' for i in cfg.device_ids]
