@@ -1,0 +1,243 @@
+#!/usr/bin/env python3
+"""Flagship benchmark: frames/sec of CLIP-ViT-B/32 ``uni_12`` extraction.
+
+Driver contract:
+  python bench.py --gpus N --steps K --warmup W
+For N > 1 the driver launches this under ``torch.distributed.run`` with one
+rank per GPU (RCCL); ranks read RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the
+environment.  Work is synthetic (no network): random-init ViT-B/32 weights,
+random uint8 224x224 RGB frames, 12 frames per "video" (the ``uni_12``
+sampling config of BASELINE.json).  Each step runs the full per-batch GPU
+pipeline — H2D copy of uint8 frames, CLIP normalization, bf16 ViT forward
+through the hand-written HIP ops (LayerNorm / QuickGELU / fused MHSA), and
+the D2H feature pull that the extractor performs per video.
+
+Weak scaling: per-GPU work is fixed (``--videos-per-step`` per rank);
+``value`` is the WHOLE-JOB frames/sec aggregated over all ranks.
+
+Secondary config (BASELINE.json config 4): ``--model i3d_raft`` measures
+clips/sec of I3D rgb+flow with RAFT flow on 64-frame stacks.
+"""
+from __future__ import annotations
+
+import argparse
+import json
+import os
+import time
+
+import torch
+
+
+def get_dist():
+    world = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    local_rank = int(os.environ.get('LOCAL_RANK', str(rank)))
+    return rank, local_rank, world
+
+
+def setup(world: int, rank: int, local_rank: int, device: torch.device):
+    if world > 1:
+        import torch.distributed as dist
+        backend = 'nccl' if device.type == 'cuda' else 'gloo'
+        os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+        os.environ.setdefault('MASTER_PORT', '29511')
+        dist.init_process_group(backend, rank=rank, world_size=world)
+
+
+def sync(device: torch.device, world: int):
+    if world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+    if device.type == 'cuda':
+        torch.cuda.synchronize(device)
+
+
+def max_over_ranks(value: float, device: torch.device, world: int) -> float:
+    if world <= 1:
+        return value
+    import torch.distributed as dist
+    t = torch.tensor([value], dtype=torch.float64,
+                     device=device if device.type == 'cuda' else 'cpu')
+    dist.all_reduce(t, op=dist.ReduceOp.MAX)
+    return float(t.item())
+
+
+# --------------------------------------------------------------------- CLIP
+def bench_clip(args, device, dtype, rank, world):
+    from video_features_amd import transforms as T
+    from video_features_amd.models.clip_vit import VisionTransformer
+    torch.manual_seed(0)
+    model = VisionTransformer().to(device=device, dtype=dtype).eval()
+    if world > 1:
+        from video_features_amd.runtime.dist import broadcast_models
+        broadcast_models(model)
+
+    frames_per_video = 12
+    videos = args.videos_per_step
+    n_frames = videos * frames_per_video
+    g = torch.Generator().manual_seed(rank + 1)
+    host_frames = torch.randint(0, 256, (n_frames, 224, 224, 3),
+                                dtype=torch.uint8, generator=g)
+    if device.type == 'cuda':
+        host_frames = host_frames.pin_memory()
+    mean = torch.tensor(T.CLIP_MEAN, device=device, dtype=dtype).view(1, 3, 1, 1)
+    std = torch.tensor(T.CLIP_STD, device=device, dtype=dtype).view(1, 3, 1, 1)
+    fb = args.frame_batch
+
+    def step():
+        outs = []
+        for s in range(0, n_frames, fb):
+            chunk = host_frames[s:s + fb].to(device, non_blocking=True)
+            x = chunk.permute(0, 3, 1, 2).to(dtype).div_(255.0)
+            x = (x - mean) / std
+            outs.append(model.encode_image(x))
+        # per-video D2H pull, as the extractor does
+        feats = torch.cat(outs).float().cpu()
+        return feats
+
+    with torch.no_grad():
+        for _ in range(args.warmup):
+            step()
+        sync(device, world)
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step()
+        sync(device, world)
+        dt = time.perf_counter() - t0
+    dt = max_over_ranks(dt, device, world)
+    total_frames = n_frames * world * args.steps
+    return {
+        'metric': 'frames/sec CLIP-ViT-B/32 uni_12',
+        'value': total_frames / dt,
+        'unit': 'frames/sec',
+        'ms_per_step': dt / args.steps * 1000.0,
+        'config': {'model': 'CLIP-ViT-B/32', 'global_batch': videos * world,
+                   'seq_len': frames_per_video, 'resolution': 224,
+                   'parallelism': f'dp{world}'},
+    }
+
+
+# ---------------------------------------------------------------- I3D+RAFT
+def bench_i3d_raft(args, device, dtype, rank, world):
+    from video_features_amd import transforms as T
+    from video_features_amd.models.i3d import I3D
+    from video_features_amd.models.raft import RAFT
+    torch.manual_seed(0)
+    i3d_rgb = I3D(modality='rgb').to(device, dtype).eval()
+    i3d_flow = I3D(modality='flow').to(device, dtype).eval()
+    raft = RAFT(iters=args.raft_iters).to(device, dtype).eval()
+    if world > 1:
+        from video_features_amd.runtime.dist import broadcast_models
+        broadcast_models({'a': i3d_rgb, 'b': i3d_flow, 'c': raft})
+
+    stack = 64
+    clips = args.clips_per_step
+    g = torch.Generator().manual_seed(rank + 1)
+    host = torch.randint(0, 256, (clips, stack + 1, 224, 224, 3),
+                         dtype=torch.uint8, generator=g)
+    if device.type == 'cuda':
+        host = host.pin_memory()
+
+    gathered = [None]
+
+    def step():
+        feats = []
+        for ci in range(clips):
+            frames = host[ci].to(device, non_blocking=True)
+            x = frames.permute(0, 3, 1, 2).to(dtype)      # (65, 3, 224, 224)
+            flow = raft(x[:-1], x[1:], test_mode=True)
+            rgb_in = T.scale_to_pm1(x[:-1]).transpose(0, 1)[None]
+            flow_in = T.i3d_flow_preprocess(flow, 224).transpose(0, 1)[None]
+            f_rgb = i3d_rgb.forward_features(rgb_in)
+            f_flow = i3d_flow.forward_features(flow_in)
+            feats.append(torch.cat([f_rgb, f_flow], dim=1))
+        out = torch.cat(feats)
+        # RCCL all-gather of stack features (BASELINE.json config 4)
+        if world > 1:
+            import torch.distributed as dist
+            buf = [torch.empty_like(out) for _ in range(world)]
+            dist.all_gather(buf, out)
+            gathered[0] = buf
+        return out.float().cpu()
+
+    with torch.no_grad():
+        for _ in range(args.warmup):
+            step()
+        sync(device, world)
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step()
+        sync(device, world)
+        dt = time.perf_counter() - t0
+    dt = max_over_ranks(dt, device, world)
+    total_clips = clips * world * args.steps
+    return {
+        'metric': 'clips/sec I3D rgb+flow (RAFT)',
+        'value': total_clips / dt,
+        'unit': 'clips/sec',
+        'ms_per_step': dt / args.steps * 1000.0,
+        'config': {'model': 'I3D+RAFT', 'global_batch': clips * world,
+                   'seq_len': stack, 'resolution': 224,
+                   'raft_iters': args.raft_iters,
+                   'parallelism': f'dp{world}'},
+    }
+
+
+def main():
+    p = argparse.ArgumentParser()
+    p.add_argument('--gpus', type=int, default=1)
+    p.add_argument('--steps', type=int, default=16)
+    p.add_argument('--warmup', type=int, default=4)
+    p.add_argument('--model', choices=['clip', 'i3d_raft'], default='clip')
+    p.add_argument('--videos-per-step', type=int, default=32,
+                   help='CLIP: synthetic videos (x12 frames) per rank per step')
+    p.add_argument('--frame-batch', type=int, default=192,
+                   help='CLIP: frames per forward chunk')
+    p.add_argument('--clips-per-step', type=int, default=2)
+    p.add_argument('--raft-iters', type=int, default=20)
+    p.add_argument('--dtype', choices=['bf16', 'fp32'], default=None)
+    args = p.parse_args()
+
+    rank, local_rank, world = get_dist()
+    if torch.cuda.is_available():
+        device = torch.device(f'cuda:{local_rank}')
+        torch.cuda.set_device(device)
+        dtype = torch.bfloat16 if args.dtype != 'fp32' else torch.float32
+    else:
+        device = torch.device('cpu')
+        dtype = torch.float32
+        args.videos_per_step = min(args.videos_per_step, 2)
+        args.clips_per_step = 1
+        args.raft_iters = min(args.raft_iters, 3)
+    setup(world, rank, local_rank, device)
+
+    if args.model == 'clip':
+        res = bench_clip(args, device, dtype, rank, world)
+    else:
+        res = bench_i3d_raft(args, device, dtype, rank, world)
+
+    if rank == 0:
+        out = {
+            'metric': res['metric'],
+            'value': round(res['value'], 2),
+            'unit': res['unit'],
+            'n_gpus': world,
+            'steps': args.steps,
+            'warmup': args.warmup,
+            'ms_per_step': round(res['ms_per_step'], 3),
+            'higher_is_better': True,
+            'scaling': 'weak',
+            'vs_baseline': None,   # the reference publishes no numbers
+            'dtype': 'bf16' if dtype == torch.bfloat16 else 'fp32',
+            'data': 'synthetic',
+            'config': res['config'],
+        }
+        print(json.dumps(out))
+
+    if world > 1:
+        import torch.distributed as dist
+        dist.destroy_process_group()
+
+
+if __name__ == '__main__':
+    main()

@@ -86,7 +86,10 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     """
     if scale is None:
         scale = 1.0 / math.sqrt(q.shape[-1])
-    if _use_hip(q) and q.dtype == torch.bfloat16 and q.shape[-1] <= 128:
+    if (_use_hip(q) and q.dtype in (torch.bfloat16, torch.float16, torch.float32)
+            and q.shape[-2] <= 64 and q.shape[-1] <= 128):
+        # fused kernel covers the ViT shapes (N<=64 tokens); longer
+        # sequences take the rocBLAS GEMM + softmax path below
         return _ext.mhsa(q.contiguous(), k.contiguous(), v.contiguous(),
                          float(scale))
     attn = (q @ k.transpose(-2, -1)) * scale

@@ -1,0 +1,142 @@
+// Python bindings for the VFA gfx950 HIP kernels (torch extension ABI).
+// Kernels live in the sibling .hip TUs behind a C ABI so this (slow,
+// torch-header-heavy) TU rarely recompiles.
+#include <torch/extension.h>
+#include <c10/hip/HIPStream.h>
+#include <hip/hip_runtime.h>
+
+extern "C" {
+void vfa_quick_gelu(const void*, void*, long long, int, hipStream_t);
+void vfa_gelu_tanh(const void*, void*, long long, int, hipStream_t);
+void vfa_layer_norm(const void*, const void*, const void*, void*, long long,
+                    int, float, int, hipStream_t);
+void vfa_bilinear_warp(const void*, const void*, void*, int, int, int, int,
+                       int, hipStream_t);
+void vfa_grid_sample(const void*, const void*, void*, long long, int, int,
+                     int, int, int, int, hipStream_t);
+void vfa_corr_repack(const void*, void*, int, int, int, int, int,
+                     hipStream_t);
+void vfa_pwc_correlation(const void*, const void*, const void*, void*, int,
+                         int, int, int, int, hipStream_t);
+void vfa_mhsa_small(const void*, const void*, const void*, void*, int, int,
+                    int, float, int, hipStream_t);
+}
+
+namespace {
+
+int dtype_tag(const torch::Tensor& t) {
+  switch (t.scalar_type()) {
+    case torch::kFloat32: return 0;
+    case torch::kBFloat16: return 1;
+    case torch::kFloat16: return 2;
+    default:
+      TORCH_CHECK(false, "unsupported dtype ", t.scalar_type());
+  }
+}
+
+hipStream_t current_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+torch::Tensor quick_gelu(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  auto out = torch::empty_like(x);
+  vfa_quick_gelu(x.data_ptr(), out.data_ptr(), x.numel(), dtype_tag(x),
+                 current_stream());
+  return out;
+}
+
+torch::Tensor gelu_tanh(torch::Tensor x) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  auto out = torch::empty_like(x);
+  vfa_gelu_tanh(x.data_ptr(), out.data_ptr(), x.numel(), dtype_tag(x),
+                current_stream());
+  return out;
+}
+
+torch::Tensor layer_norm(torch::Tensor x, torch::Tensor w, torch::Tensor b,
+                         double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  const int d = (int)x.size(-1);
+  TORCH_CHECK(w.numel() == d && b.numel() == d, "affine shape mismatch");
+  auto wc = w.contiguous().to(x.scalar_type());
+  auto bc = b.contiguous().to(x.scalar_type());
+  auto out = torch::empty_like(x);
+  vfa_layer_norm(x.data_ptr(), wc.data_ptr(), bc.data_ptr(), out.data_ptr(),
+                 x.numel() / d, d, (float)eps, dtype_tag(x), current_stream());
+  return out;
+}
+
+torch::Tensor bilinear_warp(torch::Tensor x, torch::Tensor flow) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && flow.is_contiguous());
+  TORCH_CHECK(x.dim() == 4 && flow.dim() == 4 && flow.size(1) == 2);
+  auto flow_c = flow.to(x.scalar_type()).contiguous();
+  auto out = torch::empty_like(x);
+  vfa_bilinear_warp(x.data_ptr(), flow_c.data_ptr(), out.data_ptr(),
+                    (int)x.size(0), (int)x.size(1), (int)x.size(2),
+                    (int)x.size(3), dtype_tag(x), current_stream());
+  return out;
+}
+
+torch::Tensor grid_sample_bilinear(torch::Tensor x, torch::Tensor coords) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous());
+  TORCH_CHECK(coords.dim() == 4 && coords.size(-1) == 2);
+  TORCH_CHECK(coords.size(0) == x.size(0));
+  auto cc = coords.to(x.scalar_type()).contiguous();
+  const long long n = x.size(0);
+  const int c = (int)x.size(1), h = (int)x.size(2), w = (int)x.size(3);
+  const int ho = (int)coords.size(1), wo = (int)coords.size(2);
+  auto out = torch::empty({(long)n, c, ho, wo}, x.options());
+  vfa_grid_sample(x.data_ptr(), cc.data_ptr(), out.data_ptr(), n, c, h, w, ho,
+                  wo, dtype_tag(x), current_stream());
+  return out;
+}
+
+torch::Tensor pwc_correlation(torch::Tensor f1, torch::Tensor f2,
+                              int64_t max_disp) {
+  TORCH_CHECK(f1.is_cuda() && f1.is_contiguous() && f2.is_contiguous());
+  TORCH_CHECK(max_disp == 4, "kernel is specialized for max_disp=4");
+  TORCH_CHECK(f1.sizes() == f2.sizes());
+  const int b = (int)f1.size(0), c = (int)f1.size(1);
+  const int h = (int)f1.size(2), w = (int)f1.size(3);
+  auto opts = f1.options();
+  auto f1p = torch::empty({b, h + 8, w + 8, c}, opts);
+  auto f2p = torch::empty({b, h + 8, w + 8, c}, opts);
+  const int tag = dtype_tag(f1);
+  auto stream = current_stream();
+  // tiled path (c<=64) reads f1 as NCHW directly; wave path needs both packed
+  if (c > 64) vfa_corr_repack(f1.data_ptr(), f1p.data_ptr(), b, c, h, w, tag, stream);
+  vfa_corr_repack(f2.data_ptr(), f2p.data_ptr(), b, c, h, w, tag, stream);
+  auto out = torch::empty({b, 81, h, w}, opts.dtype(torch::kFloat32));
+  vfa_pwc_correlation(f1.data_ptr(), f1p.data_ptr(), f2p.data_ptr(),
+                      out.data_ptr(), b, c, h, w, tag, stream);
+  return out.to(f1.scalar_type());
+}
+
+torch::Tensor mhsa(torch::Tensor q, torch::Tensor k, torch::Tensor v,
+                   double scale) {
+  // (B, H, N, D) each, N<=64, D<=128
+  TORCH_CHECK(q.is_cuda() && q.is_contiguous() && k.is_contiguous() &&
+              v.is_contiguous());
+  TORCH_CHECK(q.dim() == 4);
+  const int n = (int)q.size(2), d = (int)q.size(3);
+  TORCH_CHECK(n <= 64 && d <= 128, "mhsa kernel covers N<=64, D<=128");
+  const int bh = (int)(q.size(0) * q.size(1));
+  auto out = torch::empty_like(q);
+  vfa_mhsa_small(q.data_ptr(), k.data_ptr(), v.data_ptr(), out.data_ptr(), bh,
+                 n, d, (float)scale, dtype_tag(q), current_stream());
+  return out;
+}
+
+}  // namespace
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("quick_gelu", &quick_gelu);
+  m.def("gelu_tanh", &gelu_tanh);
+  m.def("layer_norm", &layer_norm);
+  m.def("bilinear_warp", &bilinear_warp);
+  m.def("grid_sample_bilinear", &grid_sample_bilinear);
+  m.def("pwc_correlation", &pwc_correlation);
+  m.def("mhsa", &mhsa);
+  m.attr("gfx_arch") = "gfx950";
+}
