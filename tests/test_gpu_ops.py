@@ -52,7 +52,7 @@ def test_gelu_tanh(dev, dtype, tol):
 
 @pytest.mark.parametrize('d', [768, 1024, 100])
 @pytest.mark.parametrize('dtype,tol', [(torch.float32, 1e-4),
-                                       (torch.bfloat16, 3e-2)])
+                                       (torch.bfloat16, 6e-2)])
 def test_layer_norm(dev, d, dtype, tol):
     ops = _hip_loaded()
     torch.manual_seed(0)

@@ -50,11 +50,8 @@ class MultiheadSelfAttention(nn.Module):
         self.proj = nn.Linear(width, width)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        b, n, e = x.shape
-        qkv = self.qkv(x).reshape(b, n, 3, self.heads, self.head_dim)
-        q, k, v = qkv.permute(2, 0, 3, 1, 4).unbind(0)   # (B, H, N, D)
-        o = ops.attention(q, k, v)
-        return self.proj(o.permute(0, 2, 1, 3).reshape(b, n, e))
+        o = ops.mhsa_fused(self.qkv(x), self.heads)
+        return self.proj(o)
 
 
 class MLP(nn.Module):

@@ -97,6 +97,27 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return attn @ v
 
 
+def mhsa_fused(qkv: torch.Tensor, heads: int,
+               scale: Optional[float] = None) -> torch.Tensor:
+    """Fused MHSA from the packed qkv projection: (B, N, 3*E) → (B, N, E).
+
+    HIP path (bf16, head_dim 64): the MFMA flash kernel consumes the packed
+    layout directly — no permute/contiguous copies on either side.  Torch
+    path: explicit reshape + softmax reference.
+    """
+    b, n, three_e = qkv.shape
+    e = three_e // 3
+    d = e // heads
+    if scale is None:
+        scale = 1.0 / math.sqrt(d)
+    if _use_hip(qkv) and qkv.dtype == torch.bfloat16 and d == 64:
+        qkv5 = qkv.view(b, n, 3, heads, d)
+        return _ext.flash_qkv(qkv5.contiguous(), float(scale))
+    q, k, v = qkv.view(b, n, 3, heads, d).permute(2, 0, 3, 1, 4).unbind(0)
+    o = attention(q.contiguous(), k.contiguous(), v.contiguous(), scale)
+    return o.permute(0, 2, 1, 3).reshape(b, n, e)
+
+
 # ---------------------------------------------------------------- flow ops
 def pwc_correlation(f1: torch.Tensor, f2: torch.Tensor,
                     max_disp: int = 4) -> torch.Tensor:

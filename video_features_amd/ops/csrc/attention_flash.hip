@@ -1,0 +1,243 @@
+// MFMA flash attention for ViT on gfx950 (bf16, head_dim = 64).
+//
+// Consumes the packed qkv projection output directly — (B, N, 3, H, 64)
+// bf16, no permute/contiguous copies — and writes (B, N, H*64) ready for
+// the output projection.  One workgroup (4 waves) per (batch*head,
+// 64-query-row block); KV tiles of 64 keys staged per tile:
+//   K as [key][d]      (row-major, +8-element row pad -> conflict-free
+//                       ds_read_b128 of 16 distinct rows, guide G4)
+//   V as [d][key]      (transposed at staging so the PV B-fragment reads
+//                       16 B contiguous)
+// Each wave owns 16 query rows:
+//   QK^T : 4x2 v_mfma_f32_16x16x32_bf16 per kv tile (A = Q frag from LDS,
+//          B = K frag from LDS, both 16 B contiguous reads)
+//   online softmax in C-fragment registers (rowmax/rowsum over the 16-lane
+//   column group via 4 shfl_xor each; m/l/O-rescale per kv tile)
+//   P -> LDS (bf16, wave-local) -> A-fragment reads; PV accumulates into
+//   the O C-fragments via MFMA with C = O.
+//
+// Fragment layouts (v_mfma_f32_16x16x32_bf16, verified by the
+// mfma_gemm16 probe + GPU numerics tests):
+//   A[i][k]: lane l holds i = l&15, k = (l>>4)*8 + j   (bf16x8)
+//   B[k][j]: lane l holds j = l&15, k = (l>>4)*8 + j'  (bf16x8)
+//   C/D[r][c]: lane l holds c = l&15, r = (l>>4)*4 + reg (f32x4)
+#include "vfa_common.h"
+
+typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
+typedef float f32x4 __attribute__((ext_vector_type(4)));
+
+namespace {
+
+constexpr int QBLK = 64;       // query rows per workgroup (16 per wave)
+constexpr int KVBLK = 64;      // keys per LDS tile
+constexpr int D = 64;          // head dim (ViT-B)
+constexpr int LSTR = D + 8;    // LDS row stride in elements (144 B, 16 B
+                               // aligned; r*36 mod 64 distinct for r<16)
+
+__device__ __forceinline__ float group16_max(float v) {
+  // reduce over the 16-lane column group (lanes differing in bits 0..3)
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) v = fmaxf(v, __shfl_xor(v, off, 64));
+  return v;
+}
+
+__device__ __forceinline__ float group16_sum(float v) {
+#pragma unroll
+  for (int off = 8; off > 0; off >>= 1) v += __shfl_xor(v, off, 64);
+  return v;
+}
+
+// stage one row-major (rows x 64) bf16 tile from global into LDS with row
+// stride LSTR; 256 threads, uint4 (8-elem) segments, coalesced.
+__device__ __forceinline__ void stage_rows(const __bf16* g, int rows,
+                                           long long row_stride, __bf16* lds,
+                                           int nvalid) {
+  const int tid = threadIdx.x;
+  for (int t = tid; t < rows * 8; t += 256) {
+    const int row = t >> 3, seg = t & 7;
+    uint4 v = {0u, 0u, 0u, 0u};
+    if (row < nvalid)
+      v = *reinterpret_cast<const uint4*>(g + row * row_stride + seg * 8);
+    *reinterpret_cast<uint4*>(lds + row * LSTR + seg * 8) = v;
+  }
+}
+
+// stage V transposed: global rows are keys (64 x 64), LDS is [d][key].
+__device__ __forceinline__ void stage_vt(const __bf16* g, long long row_stride,
+                                         __bf16* lds, int nvalid) {
+  const int tid = threadIdx.x;
+  for (int t = tid; t < KVBLK * 8; t += 256) {
+    const int key = t >> 3, seg = t & 7;
+    uint4 v = {0u, 0u, 0u, 0u};
+    if (key < nvalid)
+      v = *reinterpret_cast<const uint4*>(g + key * row_stride + seg * 8);
+    const __bf16* e = reinterpret_cast<const __bf16*>(&v);
+#pragma unroll
+    for (int j = 0; j < 8; ++j) lds[(seg * 8 + j) * LSTR + key] = e[j];
+  }
+}
+
+__global__ __launch_bounds__(256)
+void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
+                      __bf16* __restrict__ out,         // (B,N,H*D)
+                      int b_total, int n, int h_total, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* s_q = reinterpret_cast<__bf16*>(smem);            // [QBLK][LSTR]
+  __bf16* s_k = s_q + QBLK * LSTR;                          // [KVBLK][LSTR]
+  __bf16* s_vt = s_k + KVBLK * LSTR;                        // [D][LSTR]
+  __bf16* s_p = s_vt + D * LSTR;                            // [4][16][LSTR]
+
+  const int bh = blockIdx.x;
+  const int bi = bh / h_total, hi = bh % h_total;
+  const int qbase = blockIdx.y * QBLK;
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  const int lo = lane & 15, hi4 = lane >> 4;
+
+  const long long row_stride = 3LL * h_total * D;
+  const __bf16* q_g = qkv + ((long long)bi * n + qbase) * row_stride +
+                      (long long)hi * D;                    // +0 for q
+  stage_rows(q_g, QBLK, row_stride, s_q, max(0, n - qbase));
+  __syncthreads();
+
+  // per-wave state: 16 query rows [wave*16, wave*16+16)
+  f32x4 o_acc[4];   // d-tiles
+  float m_run[4], l_run[4];
+#pragma unroll
+  for (int t = 0; t < 4; ++t) o_acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+  for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
+
+  const int n_kv = (n + KVBLK - 1) / KVBLK;
+  for (int kv = 0; kv < n_kv; ++kv) {
+    const int kvbase = kv * KVBLK;
+    const int valid = min(KVBLK, n - kvbase);
+    __syncthreads();
+    const __bf16* k_g = qkv + ((long long)bi * n + kvbase) * row_stride +
+                        ((long long)1 * h_total + hi) * D;
+    const __bf16* v_g = qkv + ((long long)bi * n + kvbase) * row_stride +
+                        ((long long)2 * h_total + hi) * D;
+    stage_rows(k_g, KVBLK, row_stride, s_k, valid);
+    stage_vt(v_g, row_stride, s_vt, valid);
+    __syncthreads();
+
+    // ---- S = Q K^T for this wave's 16 rows x 64 keys
+    f32x4 s_frag[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) s_frag[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          s_q + (wave * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            s_k + (nt * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
+        s_frag[nt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, s_frag[nt], 0, 0, 0);
+      }
+    }
+
+    // ---- online softmax over keys
+    float p_vals[4][4];   // [nt][reg] exp values (bf16-packed later)
+    float alpha[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = -1e30f;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        float s = s_frag[nt][r] * scale;
+        if (kvbase + nt * 16 + lo >= n) s = -1e30f;
+        s_frag[nt][r] = s;
+        mx = fmaxf(mx, s);
+      }
+      mx = group16_max(mx);
+      const float m_new = fmaxf(m_run[r], mx);
+      alpha[r] = __expf(m_run[r] - m_new);
+      m_run[r] = m_new;
+      float rowsum = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const float p = __expf(s_frag[nt][r] - m_new);
+        p_vals[nt][r] = p;
+        rowsum += p;
+      }
+      l_run[r] = l_run[r] * alpha[r] + group16_sum(rowsum);
+    }
+    // rescale O by alpha (per row r)
+#pragma unroll
+    for (int t = 0; t < 4; ++t)
+#pragma unroll
+      for (int r = 0; r < 4; ++r) o_acc[t][r] *= alpha[r];
+
+    // ---- P through LDS (wave-local) into A-fragment layout
+    __bf16* p_lds = s_p + wave * 16 * LSTR;
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_lds[(hi4 * 4 + r) * LSTR + nt * 16 + lo] = (__bf16)p_vals[nt][r];
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    // ---- O += P V
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          p_lds + lo * LSTR + kt * 32 + hi4 * 8);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            s_vt + (nt * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
+        o_acc[nt] =
+            __builtin_amdgcn_mfma_f32_16x16x32_bf16(a, bfr, o_acc[nt], 0, 0, 0);
+      }
+    }
+  }
+
+  // ---- normalize + write (B, N, H*D)
+#pragma unroll
+  for (int r = 0; r < 4; ++r) {
+    const int grow = qbase + wave * 16 + hi4 * 4 + r;
+    if (grow >= n) continue;
+    const float inv_l = 1.0f / l_run[r];
+    __bf16* orow = out + ((long long)bi * n + grow) * (h_total * D) +
+                   (long long)hi * D;
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+      orow[nt * 16 + lo] = (__bf16)(o_acc[nt][r] * inv_l);
+  }
+}
+
+// layout probe: D(16x16) = A(16x32) @ B(32x16), row-major f32 in/out
+__global__ void mfma_gemm16_kernel(const float* a, const float* b, float* d) {
+  const int l = threadIdx.x, lo = l & 15, hi4 = l >> 4;
+  bf16x8 av, bv;
+#pragma unroll
+  for (int j = 0; j < 8; ++j) {
+    av[j] = (__bf16)a[lo * 32 + hi4 * 8 + j];
+    bv[j] = (__bf16)b[(hi4 * 8 + j) * 16 + lo];
+  }
+  f32x4 c = {0.f, 0.f, 0.f, 0.f};
+  c = __builtin_amdgcn_mfma_f32_16x16x32_bf16(av, bv, c, 0, 0, 0);
+#pragma unroll
+  for (int r = 0; r < 4; ++r) d[(hi4 * 4 + r) * 16 + lo] = c[r];
+}
+
+}  // namespace
+
+extern "C" {
+
+void vfa_flash_qkv(const void* qkv, void* out, int b, int n, int h,
+                   float scale, hipStream_t stream) {
+  dim3 grid(b * h, (n + QBLK - 1) / QBLK);
+  size_t lds = (size_t)(QBLK + KVBLK + D + 4 * 16) * LSTR * sizeof(__bf16);
+  hipLaunchKernelGGL(flash_qkv_kernel, grid, dim3(256), lds, stream,
+                     (const __bf16*)qkv, (__bf16*)out, b, n, h, scale);
+}
+
+void vfa_mfma_gemm16(const void* a, const void* b, void* d,
+                     hipStream_t stream) {
+  hipLaunchKernelGGL(mfma_gemm16_kernel, dim3(1), dim3(64), 0, stream,
+                     (const float*)a, (const float*)b, (float*)d);
+}
+
+}  // extern "C"

@@ -20,6 +20,8 @@ void vfa_pwc_correlation(const void*, const void*, const void*, void*, int,
                          int, int, int, int, hipStream_t);
 void vfa_mhsa_small(const void*, const void*, const void*, void*, int, int,
                     int, float, int, hipStream_t);
+void vfa_flash_qkv(const void*, void*, int, int, int, float, hipStream_t);
+void vfa_mfma_gemm16(const void*, const void*, void*, hipStream_t);
 }
 
 namespace {
@@ -128,6 +130,31 @@ torch::Tensor mhsa(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return out;
 }
 
+torch::Tensor flash_qkv(torch::Tensor qkv, double scale) {
+  // qkv: (B, N, 3, H, D) bf16 contiguous, D == 64 -> out (B, N, H*D)
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 5);
+  TORCH_CHECK(qkv.size(2) == 3 && qkv.size(4) == 64,
+              "flash_qkv covers head_dim 64");
+  TORCH_CHECK(qkv.scalar_type() == torch::kBFloat16);
+  const int b = (int)qkv.size(0), n = (int)qkv.size(1);
+  const int h = (int)qkv.size(3);
+  auto out = torch::empty({b, n, h * 64L}, qkv.options());
+  vfa_flash_qkv(qkv.data_ptr(), out.data_ptr(), b, n, h, (float)scale,
+                current_stream());
+  return out;
+}
+
+torch::Tensor mfma_gemm16(torch::Tensor a, torch::Tensor b) {
+  TORCH_CHECK(a.is_cuda() && a.is_contiguous() && b.is_contiguous());
+  TORCH_CHECK(a.sizes() == torch::IntArrayRef({16, 32}) &&
+              b.sizes() == torch::IntArrayRef({32, 16}));
+  auto af = a.to(torch::kFloat32), bf = b.to(torch::kFloat32);
+  auto d = torch::empty({16, 16}, af.options());
+  vfa_mfma_gemm16(af.data_ptr(), bf.data_ptr(), d.data_ptr(),
+                  current_stream());
+  return d;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -138,5 +165,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("grid_sample_bilinear", &grid_sample_bilinear);
   m.def("pwc_correlation", &pwc_correlation);
   m.def("mhsa", &mhsa);
+  m.def("flash_qkv", &flash_qkv);
+  m.def("mfma_gemm16", &mfma_gemm16);
   m.attr("gfx_arch") = "gfx950";
 }
