@@ -56,5 +56,6 @@ def test_flash_qkv_softmax_stability(dev):
     q, k, v = qkv.float().view(b, n, 3, h, d).permute(2, 0, 3, 1, 4).unbind(0)
     ref = torch.nn.functional.scaled_dot_product_attention(
         q, k, v, scale=1.0 / d ** 0.5).permute(0, 2, 1, 3).reshape(b, n, h * d)
-    err = (out - ref).abs().max().item()
-    assert err < 6e-2, err
+    # inputs are x8 scaled → compare relative to output magnitude (bf16 P·V)
+    err = ((out - ref).abs() / (ref.abs().clamp(min=1.0))).max().item()
+    assert err < 2e-2, err

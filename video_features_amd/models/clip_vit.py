@@ -96,16 +96,41 @@ class VisionTransformer(nn.Module):
         self.ln_post = LayerNorm(w)
         self.proj = nn.Parameter(scale * torch.randn(w, cfg.output_dim))
 
+    def _patch_embed(self, x: torch.Tensor) -> torch.Tensor:
+        """Patch embedding.  Stride==kernel conv IS a reshape + GEMM; on GPU
+        route it to rocBLAS directly instead of an im2col conv."""
+        if x.is_cuda:
+            t = x.shape[0]
+            p = self.cfg.patch_size
+            g = x.shape[-1] // p
+            xp = x.reshape(t, 3, g, p, g, p).permute(0, 2, 4, 1, 3, 5)
+            xp = xp.reshape(t, g * g, 3 * p * p)
+            w = self.conv1.weight.reshape(self.cfg.width, -1)
+            return xp @ w.t()                           # (T, P, W)
+        x = self.conv1(x)                               # (T, W, R/p, R/p)
+        return x.flatten(2).transpose(1, 2)             # (T, P, W)
+
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """(T, 3, R, R) preprocessed frames → (T, output_dim) features."""
-        x = self.conv1(x)                               # (T, W, R/p, R/p)
-        x = x.flatten(2).transpose(1, 2)                # (T, P, W)
+        x = self._patch_embed(x)
         cls = self.class_embedding.to(x.dtype).expand(x.shape[0], 1, -1)
         x = torch.cat([cls, x], dim=1)
         x = x + self.positional_embedding.to(x.dtype)
         x = self.ln_pre(x)
+        # residual stream carried as (delta, res): every add is fused into
+        # the next LayerNorm (ops.layer_norm_residual)
+        delta, res = None, x
         for blk in self.blocks:
-            x = blk(x)
+            if delta is None:
+                y1, s1 = blk.ln_1(res), res
+            else:
+                y1, s1 = ops.layer_norm_residual(delta, res, blk.ln_1.weight,
+                                                 blk.ln_1.bias, blk.ln_1.eps)
+            a = blk.attn(y1)
+            y2, s2 = ops.layer_norm_residual(a, s1, blk.ln_2.weight,
+                                             blk.ln_2.bias, blk.ln_2.eps)
+            delta, res = blk.mlp(y2), s2
+        x = delta + res
         x = self.ln_post(x[:, 0, :])
         return x @ self.proj.to(x.dtype)
 

@@ -97,6 +97,34 @@ def attention(q: torch.Tensor, k: torch.Tensor, v: torch.Tensor,
     return attn @ v
 
 
+def layer_norm_residual(x: torch.Tensor, res: torch.Tensor,
+                        weight: torch.Tensor, bias: torch.Tensor,
+                        eps: float = 1e-5):
+    """Fused ``s = x + res; y = layer_norm(s)`` → (y, s).  One kernel on the
+    HIP path (removes the eager add + a full re-read)."""
+    if _use_hip(x) and x.dtype in (torch.bfloat16, torch.float16, torch.float32):
+        y, s = _ext.layer_norm_residual(x.contiguous(), res.contiguous(),
+                                        weight, bias, eps)
+        return y, s
+    s = x + res
+    y = torch.nn.functional.layer_norm(s, (s.shape[-1],), weight, bias, eps)
+    return y, s
+
+
+def preprocess_u8_chw(frames_u8: torch.Tensor, mean, std,
+                      bf16: bool = True) -> torch.Tensor:
+    """(T, H, W, 3) uint8 → (T, 3, H, W) normalized bf16/f32 in one fused
+    kernel on GPU; eager chain on CPU."""
+    if _use_hip(frames_u8):
+        return _ext.u8_chw_norm(frames_u8.contiguous(), list(map(float, mean)),
+                                list(map(float, std)), bf16)
+    x = frames_u8.permute(0, 3, 1, 2).float() / 255.0
+    mean_t = torch.as_tensor(mean, dtype=x.dtype, device=x.device)
+    std_t = torch.as_tensor(std, dtype=x.dtype, device=x.device)
+    x = (x - mean_t[:, None, None]) / std_t[:, None, None]
+    return x.to(torch.bfloat16) if bf16 else x
+
+
 def mhsa_fused(qkv: torch.Tensor, heads: int,
                scale: Optional[float] = None) -> torch.Tensor:
     """Fused MHSA from the packed qkv projection: (B, N, 3*E) → (B, N, E).

@@ -22,6 +22,11 @@ void vfa_mhsa_small(const void*, const void*, const void*, void*, int, int,
                     int, float, int, hipStream_t);
 void vfa_flash_qkv(const void*, void*, int, int, int, float, hipStream_t);
 void vfa_mfma_gemm16(const void*, const void*, void*, hipStream_t);
+void vfa_layer_norm_residual(const void*, const void*, const void*,
+                             const void*, void*, void*, long long, int,
+                             float, int, hipStream_t);
+void vfa_u8_chw_norm(const void*, void*, long long, int, int, const float*,
+                     const float*, int, hipStream_t);
 }
 
 namespace {
@@ -130,6 +135,42 @@ torch::Tensor mhsa(torch::Tensor q, torch::Tensor k, torch::Tensor v,
   return out;
 }
 
+std::vector<torch::Tensor> layer_norm_residual(torch::Tensor x,
+                                               torch::Tensor res,
+                                               torch::Tensor w,
+                                               torch::Tensor b, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && res.is_contiguous());
+  TORCH_CHECK(x.sizes() == res.sizes());
+  const int d = (int)x.size(-1);
+  auto wc = w.contiguous().to(x.scalar_type());
+  auto bc = b.contiguous().to(x.scalar_type());
+  auto y = torch::empty_like(x);
+  auto s = torch::empty_like(x);
+  vfa_layer_norm_residual(x.data_ptr(), res.data_ptr(), wc.data_ptr(),
+                          bc.data_ptr(), y.data_ptr(), s.data_ptr(),
+                          x.numel() / d, d, (float)eps, dtype_tag(x),
+                          current_stream());
+  return {y, s};
+}
+
+torch::Tensor u8_chw_norm(torch::Tensor frames, std::vector<double> mean,
+                          std::vector<double> std_, bool bf16) {
+  TORCH_CHECK(frames.is_cuda() && frames.is_contiguous());
+  TORCH_CHECK(frames.scalar_type() == torch::kUInt8);
+  TORCH_CHECK(frames.dim() == 4 && frames.size(3) == 3, "(T,H,W,3) expected");
+  const long long t = frames.size(0);
+  const int h = (int)frames.size(1), w = (int)frames.size(2);
+  TORCH_CHECK((h * w) % 4 == 0);
+  float m[3] = {(float)mean[0], (float)mean[1], (float)mean[2]};
+  float s[3] = {(float)std_[0], (float)std_[1], (float)std_[2]};
+  auto out = torch::empty({(long)t, 3, h, w},
+                          frames.options().dtype(bf16 ? torch::kBFloat16
+                                                      : torch::kFloat32));
+  vfa_u8_chw_norm(frames.data_ptr(), out.data_ptr(), t, h, w, m, s,
+                  bf16 ? 1 : 0, current_stream());
+  return out;
+}
+
 torch::Tensor flash_qkv(torch::Tensor qkv, double scale) {
   // qkv: (B, N, 3, H, D) bf16 contiguous, D == 64 -> out (B, N, H*D)
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 5);
@@ -167,5 +208,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mhsa", &mhsa);
   m.def("flash_qkv", &flash_qkv);
   m.def("mfma_gemm16", &mfma_gemm16);
+  m.def("layer_norm_residual", &layer_norm_residual);
+  m.def("u8_chw_norm", &u8_chw_norm);
   m.attr("gfx_arch") = "gfx950";
 }

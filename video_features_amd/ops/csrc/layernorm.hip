@@ -77,6 +77,87 @@ __global__ void layernorm_kernel(const T* __restrict__ in,
   }
 }
 
+// residual fusion: s = x + res is written once and normalized in the same
+// launch (removes the separate eager add kernel + one full re-read)
+template <typename T>
+__global__ void layernorm_res_kernel(const T* __restrict__ x,
+                                     const T* __restrict__ res,
+                                     const T* __restrict__ weight,
+                                     const T* __restrict__ bias,
+                                     T* __restrict__ y, T* __restrict__ s_out,
+                                     int rows, int d, float eps) {
+  constexpr int VEC = 16 / sizeof(T);
+  __shared__ float red[2][8];
+  const int row = blockIdx.x;
+  if (row >= rows) return;
+  const T* xr = x + (long long)row * d;
+  const T* rr = res + (long long)row * d;
+  T* sr = s_out + (long long)row * d;
+  T* yr = y + (long long)row * d;
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int nwaves = blockDim.x >> 6;
+  using VecT = __attribute__((ext_vector_type(4))) unsigned;
+
+  float sum = 0.f, sumsq = 0.f;
+  const int dvec = d / VEC;
+  for (int i = tid; i < dvec; i += blockDim.x) {
+    T tx[VEC], tr[VEC];
+    *reinterpret_cast<VecT*>(tx) =
+        *reinterpret_cast<const VecT*>(xr + (long long)i * VEC);
+    *reinterpret_cast<VecT*>(tr) =
+        *reinterpret_cast<const VecT*>(rr + (long long)i * VEC);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v = to_f32<T>(tx[j]) + to_f32<T>(tr[j]);
+      tx[j] = from_f32<T>(v);
+      sum += v;
+      sumsq += v * v;
+    }
+    *reinterpret_cast<VecT*>(sr + (long long)i * VEC) =
+        *reinterpret_cast<VecT*>(tx);
+  }
+  for (int i = dvec * VEC + tid; i < d; i += blockDim.x) {
+    float v = to_f32<T>(xr[i]) + to_f32<T>(rr[i]);
+    sr[i] = from_f32<T>(v);
+    sum += v;
+    sumsq += v * v;
+  }
+  sum = wave_reduce_sum(sum);
+  sumsq = wave_reduce_sum(sumsq);
+  if (lane == 0) { red[0][wave] = sum; red[1][wave] = sumsq; }
+  __syncthreads();
+  if (tid == 0) {
+    float s = 0.f, ss = 0.f;
+    for (int w2 = 0; w2 < nwaves; ++w2) { s += red[0][w2]; ss += red[1][w2]; }
+    float mean = s / d;
+    red[0][0] = mean;
+    red[1][0] = rsqrtf(ss / d - mean * mean + eps);
+  }
+  __syncthreads();
+  const float mean = red[0][0], rstd = red[1][0];
+  for (int i = tid; i < dvec; i += blockDim.x) {
+    T tx[VEC], tw[VEC], tb[VEC];
+    *reinterpret_cast<VecT*>(tx) =
+        *reinterpret_cast<const VecT*>(sr + (long long)i * VEC);
+    *reinterpret_cast<VecT*>(tw) =
+        *reinterpret_cast<const VecT*>(weight + (long long)i * VEC);
+    *reinterpret_cast<VecT*>(tb) =
+        *reinterpret_cast<const VecT*>(bias + (long long)i * VEC);
+#pragma unroll
+    for (int j = 0; j < VEC; ++j) {
+      float v = (to_f32<T>(tx[j]) - mean) * rstd;
+      tx[j] = from_f32<T>(v * to_f32<T>(tw[j]) + to_f32<T>(tb[j]));
+    }
+    *reinterpret_cast<VecT*>(yr + (long long)i * VEC) =
+        *reinterpret_cast<VecT*>(tx);
+  }
+  for (int i = dvec * VEC + tid; i < d; i += blockDim.x) {
+    float v = (to_f32<T>(sr[i]) - mean) * rstd;
+    yr[i] = from_f32<T>(v * to_f32<T>(weight[i]) + to_f32<T>(bias[i]));
+  }
+}
+
 template <typename T>
 void launch_ln(const void* in, const void* w, const void* b, void* out,
                long long rows, int d, float eps, hipStream_t stream) {
@@ -96,5 +177,36 @@ extern "C" void vfa_layer_norm(const void* in, const void* w, const void* b,
     case VFA_BF16:
       launch_ln<__hip_bfloat16>(in, w, b, out, rows, d, eps, stream); break;
     case VFA_F16: launch_ln<__half>(in, w, b, out, rows, d, eps, stream); break;
+  }
+}
+
+extern "C" void vfa_layer_norm_residual(const void* x, const void* res,
+                                        const void* w, const void* b,
+                                        void* y, void* s_out, long long rows,
+                                        int d, float eps, int dtype,
+                                        hipStream_t stream) {
+  int block = d >= 2048 ? 512 : 256;
+  switch (dtype) {
+    case VFA_F32:
+      hipLaunchKernelGGL((layernorm_res_kernel<float>), dim3((unsigned)rows),
+                         dim3(block), 0, stream, (const float*)x,
+                         (const float*)res, (const float*)w, (const float*)b,
+                         (float*)y, (float*)s_out, (int)rows, d, eps);
+      break;
+    case VFA_BF16:
+      hipLaunchKernelGGL((layernorm_res_kernel<__hip_bfloat16>),
+                         dim3((unsigned)rows), dim3(block), 0, stream,
+                         (const __hip_bfloat16*)x, (const __hip_bfloat16*)res,
+                         (const __hip_bfloat16*)w, (const __hip_bfloat16*)b,
+                         (__hip_bfloat16*)y, (__hip_bfloat16*)s_out,
+                         (int)rows, d, eps);
+      break;
+    case VFA_F16:
+      hipLaunchKernelGGL((layernorm_res_kernel<__half>), dim3((unsigned)rows),
+                         dim3(block), 0, stream, (const __half*)x,
+                         (const __half*)res, (const __half*)w,
+                         (const __half*)b, (__half*)y, (__half*)s_out,
+                         (int)rows, d, eps);
+      break;
   }
 }
