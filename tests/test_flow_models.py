@@ -98,3 +98,26 @@ def test_flow_extractor_end_to_end(y4m_video, feature_type):
     flow = out[feature_type]
     assert flow.shape == (15, 2, 64, 96)   # T-1 flow frames
     assert np.isfinite(flow).all()
+
+
+def test_raft_checkpoint_compat_separate_zr():
+    """Reference-style checkpoints with separate convz/convr keys load into
+    the merged convzr layout."""
+    from video_features_amd.models.raft import SepConvGRU
+    torch.manual_seed(0)
+    gru = SepConvGRU(8, 16)
+    sd = gru.state_dict()
+    legacy = {}
+    for i in ('1', '2'):
+        w = sd[f'convzr{i}.weight']
+        bi = sd[f'convzr{i}.bias']
+        legacy[f'convz{i}.weight'], legacy[f'convr{i}.weight'] = \
+            w[:8].clone(), w[8:].clone()
+        legacy[f'convz{i}.bias'], legacy[f'convr{i}.bias'] = \
+            bi[:8].clone(), bi[8:].clone()
+        legacy[f'convq{i}.weight'] = sd[f'convq{i}.weight'].clone()
+        legacy[f'convq{i}.bias'] = sd[f'convq{i}.bias'].clone()
+    gru2 = SepConvGRU(8, 16)
+    gru2.load_state_dict(legacy)
+    for k, v in gru2.state_dict().items():
+        assert torch.equal(v, sd[k]), k

@@ -51,13 +51,21 @@ class Unit3D(nn.Module):
         super().__init__()
         self.kernel = _triple(kernel)
         self.stride = _triple(stride)
+        # stride-1 + odd kernel ⇒ TF-SAME total pad k-1 splits symmetrically,
+        # so the conv's own padding is exact and the F.pad copy is skipped
+        # (true for every I3D conv except the 7×7×7/2 stem)
+        self.static_same = all(s == 1 for s in self.stride) and \
+            all(k % 2 == 1 for k in self.kernel)
+        pad = tuple(k // 2 for k in self.kernel) if self.static_same else 0
         self.conv = nn.Conv3d(in_ch, out_ch, self.kernel, self.stride,
-                              padding=0, bias=use_bias)
+                              padding=pad, bias=use_bias)
         self.bn = nn.BatchNorm3d(out_ch, eps=1e-3) if use_bn else None
         self.activation = activation
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.conv(tf_same_pad_3d(x, self.kernel, self.stride))
+        if not self.static_same:
+            x = tf_same_pad_3d(x, self.kernel, self.stride)
+        x = self.conv(x)
         if self.bn is not None:
             x = self.bn(x)
         return F.relu(x, inplace=True) if self.activation else x

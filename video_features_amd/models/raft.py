@@ -5,10 +5,22 @@ Re-implementation of the network the reference vendors
 stride-8 feature/context encoders, all-pairs 4-level correlation pyramid,
 iterative SepConvGRU updates, convex-combination 8× upsampling.
 
-MI355X mapping: the all-pairs correlation is a plain batched GEMM
-(rocBLAS via torch.matmul), while the per-iteration pyramid *lookup* — a
-bilinear gather of 4×(2r+1)² taps per pixel — dispatches through
-``ops.grid_sample_bilinear`` (hand-written HIP gather kernel on GPU).
+MI355X mapping (all decisions measured with rocprofv3, profiles/):
+  * all-pairs correlation = one rocBLAS batched GEMM;
+  * the per-iteration pyramid lookup (4 levels × 81 taps) is ONE fused HIP
+    kernel (``ops.corr_lookup``) with the per-pixel correlation planes
+    staged through LDS — the reference runs 4 grid_samples plus window
+    tensor construction per iteration;
+  * the SepConvGRU z/r convs are merged into one conv and the gate
+    elementwise is fused (``ops.gru_zr`` / ``ops.gru_out``), with the
+    conv inputs kept in two persistent (B, 384, H/8, W/8) buffers instead
+    of 4 torch.cat allocations per iteration;
+  * the convex 8× upsample is one kernel (``ops.convex_upsample``) and —
+    like the reference's test mode, which only returns the final flow
+    (reference raft.py:170-172) — is computed on the LAST iteration only;
+  * ``use_channels_last()`` flips the whole module to NHWC so MIOpen picks
+    its NHWC igemm solvers without batched_transpose fixups; the fused
+    kernels take an nhwc flag and address accordingly.
 """
 from __future__ import annotations
 
@@ -88,32 +100,22 @@ class CorrPyramid:
         self.num_levels = num_levels
         self.radius = radius
         b, d, h, w = fmap1.shape
-        f1 = fmap1.flatten(2).transpose(1, 2)        # (B, HW, D)
-        f2 = fmap2.flatten(2)                        # (B, D, HW)
+        f1 = fmap1.reshape(b, d, h * w).transpose(1, 2)  # (B, HW, D)
+        f2 = fmap2.reshape(b, d, h * w)                  # (B, D, HW)
         corr = torch.matmul(f1, f2) / (d ** 0.5)     # rocBLAS batched GEMM
         corr = corr.reshape(b * h * w, 1, h, w)
         self.shape = (b, h, w)
-        self.pyramid: List[torch.Tensor] = [corr]
+        self.pyramid: List[torch.Tensor] = [corr.contiguous()]
         for _ in range(num_levels - 1):
             corr = F.avg_pool2d(corr, 2, 2)
-            self.pyramid.append(corr)
+            self.pyramid.append(corr.contiguous())
 
-    def __call__(self, coords: torch.Tensor) -> torch.Tensor:
-        """coords (B, 2, H, W) in pixels at 1/8 res → (B, L*(2r+1)^2, H, W)."""
-        r = self.radius
-        b, h, w = self.shape
-        coords = coords.permute(0, 2, 3, 1)          # (B, H, W, 2)
-        out = []
-        for lvl, corr in enumerate(self.pyramid):
-            dx = torch.linspace(-r, r, 2 * r + 1, device=coords.device,
-                                dtype=coords.dtype)
-            delta = torch.stack(torch.meshgrid(dx, dx, indexing='ij'),
-                                dim=-1).flip(-1)     # (2r+1, 2r+1, 2) xy order
-            centroid = coords.reshape(b * h * w, 1, 1, 2) / (2 ** lvl)
-            window = centroid + delta[None]
-            sampled = ops.grid_sample_bilinear(corr, window)
-            out.append(sampled.reshape(b, h, w, -1))
-        return torch.cat(out, dim=-1).permute(0, 3, 1, 2).contiguous()
+    def __call__(self, coords: torch.Tensor, nhwc: bool = False,
+                 out_dtype: torch.dtype = None) -> torch.Tensor:
+        """coords (B, 2, H, W) fp32 pixels at 1/8 res →
+        (B, L*(2r+1)^2, H, W)."""
+        return ops.corr_lookup(self.pyramid, coords.contiguous(),
+                               self.radius, nhwc, out_dtype)
 
 
 # ----------------------------------------------------------------- update
@@ -128,29 +130,43 @@ class FlowHead(nn.Module):
 
 
 class SepConvGRU(nn.Module):
-    """Separable 1×5 / 5×1 ConvGRU (reference update.py:37-64)."""
+    """Separable 1×5 / 5×1 ConvGRU (reference update.py:37-64), with the
+    z and r convs merged into one ``convzr`` and the gate elementwise fused.
+
+    ``forward`` operates on two persistent conv-input buffers ``hx``/``rhx``
+    of shape (B, hidden+in_dim, H, W): channels [0, hidden) hold h (updated
+    in place by ``ops.gru_out``), channels [hidden, ...) hold x (written
+    once per RAFT iteration by the caller).
+    """
 
     def __init__(self, hidden: int = 128, in_dim: int = 256):
         super().__init__()
+        self.hidden = hidden
         c = hidden + in_dim
-        self.convz1 = nn.Conv2d(c, hidden, (1, 5), padding=(0, 2))
-        self.convr1 = nn.Conv2d(c, hidden, (1, 5), padding=(0, 2))
+        self.convzr1 = nn.Conv2d(c, 2 * hidden, (1, 5), padding=(0, 2))
         self.convq1 = nn.Conv2d(c, hidden, (1, 5), padding=(0, 2))
-        self.convz2 = nn.Conv2d(c, hidden, (5, 1), padding=(2, 0))
-        self.convr2 = nn.Conv2d(c, hidden, (5, 1), padding=(2, 0))
+        self.convzr2 = nn.Conv2d(c, 2 * hidden, (5, 1), padding=(2, 0))
         self.convq2 = nn.Conv2d(c, hidden, (5, 1), padding=(2, 0))
 
-    def _step(self, h, x, convz, convr, convq):
-        hx = torch.cat([h, x], dim=1)
-        z = torch.sigmoid(convz(hx))
-        r = torch.sigmoid(convr(hx))
-        q = torch.tanh(convq(torch.cat([r * h, x], dim=1)))
-        return (1 - z) * h + z * q
+    def _load_from_state_dict(self, state_dict, prefix, *args, **kwargs):
+        # accept reference-style checkpoints with separate convz/convr
+        for i in ('1', '2'):
+            zw = prefix + f'convz{i}.weight'
+            rw = prefix + f'convr{i}.weight'
+            if zw in state_dict and rw in state_dict:
+                state_dict[prefix + f'convzr{i}.weight'] = torch.cat(
+                    [state_dict.pop(zw), state_dict.pop(rw)])
+                state_dict[prefix + f'convzr{i}.bias'] = torch.cat(
+                    [state_dict.pop(prefix + f'convz{i}.bias'),
+                     state_dict.pop(prefix + f'convr{i}.bias')])
+        super()._load_from_state_dict(state_dict, prefix, *args, **kwargs)
 
-    def forward(self, h, x):
-        h = self._step(h, x, self.convz1, self.convr1, self.convq1)
-        h = self._step(h, x, self.convz2, self.convr2, self.convq2)
-        return h
+    def forward(self, hx: torch.Tensor, rhx: torch.Tensor,
+                nhwc: bool = False) -> None:
+        z = ops.gru_zr(self.convzr1(hx), hx, rhx, nhwc)
+        ops.gru_out(self.convq1(rhx), z, hx, nhwc)
+        z = ops.gru_zr(self.convzr2(hx), hx, rhx, nhwc)
+        ops.gru_out(self.convq2(rhx), z, hx, nhwc)
 
 
 class BasicMotionEncoder(nn.Module):
@@ -178,16 +194,11 @@ class BasicUpdateBlock(nn.Module):
         self.encoder = BasicMotionEncoder()
         self.gru = SepConvGRU(hidden, in_dim=128 + context)
         self.flow_head = FlowHead(hidden, 256)
+        # raw logits; the 0.25 scale of reference raft.py:158 is folded
+        # into ops.convex_upsample
         self.mask = nn.Sequential(nn.Conv2d(hidden, 256, 3, 1, 1),
                                   nn.ReLU(inplace=True),
                                   nn.Conv2d(256, 64 * 9, 1))
-
-    def forward(self, net, inp, corr, flow):
-        motion = self.encoder(flow, corr)
-        net = self.gru(net, torch.cat([inp, motion], dim=1))
-        delta_flow = self.flow_head(net)
-        up_mask = 0.25 * self.mask(net)
-        return net, up_mask, delta_flow
 
 
 # ------------------------------------------------------------------- RAFT
@@ -197,9 +208,15 @@ class RAFT(nn.Module):
         super().__init__()
         self.hdim, self.cdim = hidden_dim, context_dim
         self.iters = iters
+        self.nhwc = False
         self.fnet = BasicEncoder(256, 'instance')
         self.cnet = BasicEncoder(hidden_dim + context_dim, 'batch')
         self.update_block = BasicUpdateBlock(hidden_dim, context_dim)
+
+    def use_channels_last(self) -> 'RAFT':
+        """Switch the conv path to NHWC (MIOpen igemm without transposes)."""
+        self.nhwc = True
+        return self.to(memory_format=torch.channels_last)
 
     @staticmethod
     def coords_grid(b: int, h: int, w: int, device, dtype):
@@ -208,40 +225,56 @@ class RAFT(nn.Module):
                                 indexing='ij')
         return torch.stack([xx, yy])[None].expand(b, -1, -1, -1).contiguous()
 
-    @staticmethod
-    def upsample_flow(flow: torch.Tensor, mask: torch.Tensor) -> torch.Tensor:
-        """Convex-combination 8× upsample (reference raft.py:100-111)."""
-        b, _, h, w = flow.shape
-        mask = mask.view(b, 1, 9, 8, 8, h, w).softmax(dim=2)
-        up = F.unfold(8 * flow, 3, padding=1).view(b, 2, 9, 1, 1, h, w)
-        up = (mask * up).sum(dim=2)                 # (B, 2, 8, 8, H, W)
-        return up.permute(0, 1, 4, 2, 5, 3).reshape(b, 2, 8 * h, 8 * w)
-
     def forward(self, image1: torch.Tensor, image2: torch.Tensor,
                 iters: int = None, test_mode: bool = True):
         """uint8-range (B, 3, H, W) pairs (H, W divisible by 8) → (B, 2, H, W)
         flow (reference raft.py:113-174)."""
         iters = iters or self.iters
-        image1 = 2 * (image1 / 255.0) - 1.0
-        image2 = 2 * (image2 / 255.0) - 1.0
-        fmap1 = self.fnet(image1)
-        fmap2 = self.fnet(image2)
+        dtype = self.fnet.conv1.weight.dtype
+        mf = torch.channels_last if self.nhwc else torch.contiguous_format
+        image1 = 2 * (image1.to(dtype) / 255.0) - 1.0
+        image2 = 2 * (image2.to(dtype) / 255.0) - 1.0
+        image1 = image1.contiguous(memory_format=mf)
+        image2 = image2.contiguous(memory_format=mf)
+        b = image1.shape[0]
+        # one batched encoder pass over both images (fewer, larger GEMMs)
+        fmap = self.fnet(torch.cat([image1, image2]))
+        fmap1, fmap2 = fmap[:b], fmap[b:]
         corr_fn = CorrPyramid(fmap1.float(), fmap2.float())
         cnet = self.cnet(image1)
         net, inp = torch.split(cnet, [self.hdim, self.cdim], dim=1)
-        net = torch.tanh(net)
         inp = F.relu(inp)
-        b, _, h8, w8 = fmap1.shape
-        coords0 = self.coords_grid(b, h8, w8, fmap1.device, torch.float32)
+        _, _, h8, w8 = fmap1.shape
+
+        # persistent GRU conv-input buffers: [h | x]
+        cbuf = self.hdim + 128 + self.cdim
+        hx = torch.empty(b, cbuf, h8, w8, device=image1.device, dtype=dtype,
+                         memory_format=mf)
+        rhx = torch.empty_like(hx, memory_format=mf)
+        hx[:, :self.hdim] = torch.tanh(net)
+
+        coords0 = self.coords_grid(b, h8, w8, image1.device, torch.float32)
         coords1 = coords0.clone()
+        ub = self.update_block
         flow_up = None
-        for _ in range(iters):
-            corr = corr_fn(coords1).to(net.dtype)
-            flow = (coords1 - coords0).to(net.dtype)
-            net, up_mask, delta = self.update_block(net, inp, corr, flow)
-            coords1 = coords1 + delta.float()
-            flow_up = self.upsample_flow((coords1 - coords0).to(net.dtype),
-                                         up_mask)
+        for it in range(iters):
+            corr = corr_fn(coords1, nhwc=self.nhwc, out_dtype=dtype)
+            flow = (coords1 - coords0).to(dtype).contiguous(memory_format=mf)
+            motion = ub.encoder(flow, corr)
+            x = torch.cat([inp, motion], dim=1)
+            hx[:, self.hdim:] = x
+            rhx[:, self.hdim:] = x
+            ub.gru(hx, rhx, self.nhwc)
+            net = hx[:, :self.hdim]
+            delta = ub.flow_head(net)
+            coords1 = (coords1 + delta.float()).contiguous()
+            if it == iters - 1 or not test_mode:
+                # test mode only needs the final upsampled flow
+                # (reference raft.py:170-172)
+                up_mask = ub.mask(net)
+                fl = (coords1 - coords0).to(dtype).contiguous(
+                    memory_format=mf)
+                flow_up = ops.convex_upsample(fl, up_mask, self.nhwc)
         if test_mode:
             return flow_up
         return coords1 - coords0, flow_up

@@ -197,6 +197,89 @@ def bilinear_warp(x: torch.Tensor, flow: torch.Tensor) -> torch.Tensor:
     return warped * (mask > 0.999).to(x.dtype)
 
 
+def corr_lookup(pyramid, coords: torch.Tensor, radius: int = 4,
+                nhwc: bool = False,
+                out_dtype: Optional[torch.dtype] = None) -> torch.Tensor:
+    """Fused RAFT correlation-pyramid lookup.
+
+    ``pyramid``: list of up to 4 levels, each (B*H*W, 1, h_l, w_l) fp32
+    (one correlation plane per query pixel, as built by CorrPyramid);
+    ``coords``: (B, 2, H, W) fp32 pixel coords at level 0.  Returns
+    (B, L*(2r+1)^2, H, W) — the concatenation over levels of the bilinear
+    samples of the (2r+1)^2 displacement window, matching the reference
+    lookup (models/raft/raft_src/corr.py:36-50) but as ONE kernel with the
+    per-pixel planes staged through LDS.
+    """
+    if out_dtype is None:
+        out_dtype = coords.dtype
+    if _use_hip(coords) and radius == 4:
+        return _ext.corr_lookup(list(pyramid), coords.contiguous(), nhwc,
+                                out_dtype)
+    # torch reference: per-level grid_sample of the displacement window
+    b, _, h, w = coords.shape
+    r = radius
+    cc = coords.permute(0, 2, 3, 1)
+    out = []
+    for lvl, corr in enumerate(pyramid):
+        dx = torch.linspace(-r, r, 2 * r + 1, device=coords.device,
+                            dtype=torch.float32)
+        delta = torch.stack(torch.meshgrid(dx, dx, indexing='ij'),
+                            dim=-1).flip(-1)
+        centroid = cc.reshape(b * h * w, 1, 1, 2) / (2 ** lvl)
+        window = centroid + delta[None]
+        sampled = grid_sample_bilinear(corr, window)
+        out.append(sampled.reshape(b, h, w, -1))
+    res = torch.cat(out, dim=-1).permute(0, 3, 1, 2).to(out_dtype)
+    return res.contiguous(memory_format=torch.channels_last) if nhwc \
+        else res.contiguous()
+
+
+def convex_upsample(flow: torch.Tensor, mask: torch.Tensor,
+                    nhwc: bool = False) -> torch.Tensor:
+    """RAFT convex-combination 8x flow upsample (reference raft.py:100-111).
+
+    ``flow`` (B, 2, h, w); ``mask`` (B, 576, h, w) RAW conv output — the
+    0.25 scale and the softmax over the 9 neighbors happen inside.  Returns
+    (B, 2, 8h, 8w) contiguous.  One kernel on GPU (replaces unfold + softmax
+    + mul-sum + two permutes).
+    """
+    if _use_hip(flow):
+        return _ext.convex_upsample(flow, mask, nhwc)
+    b, _, h, w = flow.shape
+    m = (0.25 * mask.float()).view(b, 1, 9, 8, 8, h, w).softmax(dim=2)
+    up = torch.nn.functional.unfold(8 * flow.float(), 3, padding=1)
+    up = up.view(b, 2, 9, 1, 1, h, w)
+    up = (m * up).sum(dim=2)
+    return up.permute(0, 1, 4, 2, 5, 3).reshape(b, 2, 8 * h, 8 * w) \
+        .to(flow.dtype)
+
+
+def gru_zr(zr: torch.Tensor, hx: torch.Tensor, rhx: torch.Tensor,
+           nhwc: bool = False) -> torch.Tensor:
+    """Fused SepConvGRU z/r gates: ``zr`` (B, 2C, h, w) is the merged
+    z|r conv output; ``hx``/``rhx`` are the persistent (B, C+X, h, w)
+    conv-input buffers whose first C channels hold h.  Computes
+    z = sigmoid(zr[:, :C]), writes sigmoid(zr[:, C:]) * h into rhx[:, :C]
+    in place, returns z."""
+    c = zr.shape[1] // 2
+    if _use_hip(zr):
+        return _ext.gru_zr(zr, hx, rhx, nhwc)
+    z = torch.sigmoid(zr[:, :c])
+    r = torch.sigmoid(zr[:, c:])
+    rhx[:, :c] = r * hx[:, :c]
+    return z
+
+
+def gru_out(q: torch.Tensor, z: torch.Tensor, hx: torch.Tensor,
+            nhwc: bool = False) -> None:
+    """Fused GRU update: hx[:, :C] = (1-z)*hx[:, :C] + z*tanh(q), in place."""
+    if _use_hip(q):
+        _ext.gru_out(q, z, hx, nhwc)
+        return
+    c = q.shape[1]
+    hx[:, :c] = (1 - z) * hx[:, :c] + z * torch.tanh(q)
+
+
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:
     """RAFT-style bilinear lookup: ``coords`` (B, Ho, Wo, 2) in *pixel* units,
     zero padding outside (reference models/raft/raft_src/utils/utils.py:57-71)."""

@@ -27,6 +27,15 @@ void vfa_layer_norm_residual(const void*, const void*, const void*,
                              float, int, hipStream_t);
 void vfa_u8_chw_norm(const void*, void*, long long, int, int, const float*,
                      const float*, int, hipStream_t);
+void vfa_corr_lookup(const void*, const void*, const void*, const void*,
+                     const void*, void*, int, int, int, int, int, int, int,
+                     int, int, long long, int, int, int, int, hipStream_t);
+void vfa_convex_upsample(const void*, const void*, void*, int, int, int, int,
+                         int, hipStream_t);
+void vfa_gru_zr(const void*, const void*, void*, void*, int, int, int,
+                long long, int, int, hipStream_t);
+void vfa_gru_out(const void*, const void*, void*, int, int, int, long long,
+                 int, int, hipStream_t);
 }
 
 namespace {
@@ -196,6 +205,108 @@ torch::Tensor mfma_gemm16(torch::Tensor a, torch::Tensor b) {
   return d;
 }
 
+torch::Tensor corr_lookup(std::vector<torch::Tensor> pyramid,
+                          torch::Tensor coords, bool nhwc,
+                          torch::ScalarType out_dtype) {
+  // pyramid: up to 4 levels of (N, 1, h_l, w_l) fp32, N = B*H*W; coords
+  // (B, 2, H, W) fp32 pixel units at level 0.  Returns (B, L*81, H, W)
+  // (contiguous, or channels_last when nhwc) in out_dtype.
+  const int levels = (int)pyramid.size();
+  TORCH_CHECK(levels >= 1 && levels <= 4, "1..4 pyramid levels");
+  TORCH_CHECK(coords.is_cuda() && coords.dim() == 4 && coords.size(1) == 2);
+  auto cc = coords.to(torch::kFloat32).contiguous();
+  const long long b = coords.size(0);
+  const int h = (int)coords.size(2), w = (int)coords.size(3);
+  const long long npix = b * h * w;
+  int hs[4] = {1, 1, 1, 1}, ws[4] = {1, 1, 1, 1};
+  const void* ptrs[4] = {nullptr, nullptr, nullptr, nullptr};
+  long long lds_floats = 0;
+  for (int l = 0; l < levels; ++l) {
+    auto& t = pyramid[l];
+    TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+                t.scalar_type() == torch::kFloat32);
+    TORCH_CHECK(t.size(0) == npix && t.size(1) == 1, "plane-per-pixel");
+    hs[l] = (int)t.size(2);
+    ws[l] = (int)t.size(3);
+    ptrs[l] = t.data_ptr();
+    lds_floats += hs[l] * ws[l];
+  }
+  for (int l = levels; l < 4; ++l) ptrs[l] = ptrs[levels - 1];
+  const int ctot = levels * 81;
+  auto opts = coords.options().dtype(out_dtype);
+  torch::Tensor out;
+  if (nhwc) {
+    out = torch::empty({b, h, w, ctot}, opts)
+              .permute({0, 3, 1, 2});  // memory is NHWC; view as NCHW
+  } else {
+    out = torch::empty({b, ctot, h, w}, opts);
+  }
+  int tag = out_dtype == torch::kFloat32 ? 0
+            : out_dtype == torch::kBFloat16 ? 1 : 2;
+  vfa_corr_lookup(ptrs[0], ptrs[1], ptrs[2], ptrs[3], cc.data_ptr(),
+                  out.data_ptr(), levels, hs[0], ws[0], hs[1], ws[1], hs[2],
+                  ws[2], hs[3], ws[3], npix, h * w,
+                  (int)(lds_floats <= 16384 ? lds_floats : 0), nhwc ? 1 : 0,
+                  tag, current_stream());
+  return out;
+}
+
+static bool cl_contig(const torch::Tensor& t) {
+  return t.is_contiguous(torch::MemoryFormat::ChannelsLast);
+}
+
+torch::Tensor convex_upsample(torch::Tensor flow, torch::Tensor mask,
+                              bool nhwc) {
+  TORCH_CHECK(flow.is_cuda() && flow.dim() == 4 && flow.size(1) == 2);
+  TORCH_CHECK(mask.dim() == 4 && mask.size(1) == 576);
+  TORCH_CHECK(mask.scalar_type() == flow.scalar_type());
+  if (nhwc) {
+    TORCH_CHECK(cl_contig(flow) && cl_contig(mask), "channels_last expected");
+  } else {
+    TORCH_CHECK(flow.is_contiguous() && mask.is_contiguous());
+  }
+  const int b = (int)flow.size(0), h = (int)flow.size(2),
+            w = (int)flow.size(3);
+  auto out = torch::empty({b, 2, 8 * h, 8 * w}, flow.options());
+  vfa_convex_upsample(flow.data_ptr(), mask.data_ptr(), out.data_ptr(), b, h,
+                      w, nhwc ? 1 : 0, dtype_tag(flow), current_stream());
+  return out;
+}
+
+torch::Tensor gru_zr(torch::Tensor zr, torch::Tensor hx, torch::Tensor rhx,
+                     bool nhwc) {
+  // zr (B,2C,h,w) conv out; hx/rhx (B,C+X,h,w) persistent buffers.
+  // Writes r*sigmoid into rhx[:, :C]; returns z (B,C,h,w).
+  const int b = (int)zr.size(0), c2 = (int)zr.size(1);
+  const int c = c2 / 2, ctot = (int)hx.size(1);
+  const long long hw = (long long)zr.size(2) * zr.size(3);
+  if (nhwc) {
+    TORCH_CHECK(cl_contig(zr) && cl_contig(hx) && cl_contig(rhx));
+  } else {
+    TORCH_CHECK(zr.is_contiguous() && hx.is_contiguous() &&
+                rhx.is_contiguous());
+  }
+  auto z = nhwc ? torch::empty({b, (int)zr.size(2), (int)zr.size(3), c},
+                               zr.options()).permute({0, 3, 1, 2})
+                : torch::empty({b, c, (int)zr.size(2), (int)zr.size(3)},
+                               zr.options());
+  vfa_gru_zr(zr.data_ptr(), hx.data_ptr(), rhx.data_ptr(), z.data_ptr(), b,
+             c, ctot - c, hw, nhwc ? 1 : 0, dtype_tag(zr), current_stream());
+  return z;
+}
+
+void gru_out(torch::Tensor q, torch::Tensor z, torch::Tensor hx, bool nhwc) {
+  const int b = (int)q.size(0), c = (int)q.size(1), ctot = (int)hx.size(1);
+  const long long hw = (long long)q.size(2) * q.size(3);
+  if (nhwc) {
+    TORCH_CHECK(cl_contig(q) && cl_contig(hx));
+  } else {
+    TORCH_CHECK(q.is_contiguous() && hx.is_contiguous());
+  }
+  vfa_gru_out(q.data_ptr(), z.data_ptr(), hx.data_ptr(), b, c, ctot - c, hw,
+              nhwc ? 1 : 0, dtype_tag(q), current_stream());
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -210,5 +321,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("mfma_gemm16", &mfma_gemm16);
   m.def("layer_norm_residual", &layer_norm_residual);
   m.def("u8_chw_norm", &u8_chw_norm);
+  m.def("corr_lookup", &corr_lookup);
+  m.def("convex_upsample", &convex_upsample);
+  m.def("gru_zr", &gru_zr);
+  m.def("gru_out", &gru_out);
   m.attr("gfx_arch") = "gfx950";
 }
