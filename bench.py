@@ -246,12 +246,141 @@ def bench_i3d_raft(args, device, dtype, rank, world):
     }
 
 
+# ---------------------------------------------------------------- ResNet-50
+def bench_resnet(args, device, dtype, rank, world):
+    """BASELINE.json config 3: ResNet-50 fix_2 frames/sec, DP-sharded."""
+    from video_features_amd import transforms as T
+    from video_features_amd.models.resnet import build_resnet
+    from video_features_amd import ops
+    torch.manual_seed(0)
+    model = build_resnet('resnet50').to(device, dtype).eval()
+    if device.type == 'cuda':
+        model = model.to(memory_format=torch.channels_last)
+    if world > 1:
+        from video_features_amd.runtime.dist import broadcast_models
+        broadcast_models(model)
+
+    n_frames = args.videos_per_step * 12
+    g = torch.Generator().manual_seed(rank + 1)
+    host = torch.randint(0, 256, (n_frames, 224, 224, 3), dtype=torch.uint8,
+                         generator=g)
+    if device.type == 'cuda':
+        host = host.pin_memory()
+    fb = min(args.frame_batch, n_frames)
+    bf16 = dtype == torch.bfloat16
+
+    def fwd(frames_u8):
+        x = ops.preprocess_u8_chw(frames_u8, T.IMAGENET_MEAN, T.IMAGENET_STD,
+                                  bf16).to(dtype)
+        if device.type == 'cuda':
+            x = x.contiguous(memory_format=torch.channels_last)
+        return model.forward_features(x)
+
+    feats = torch.empty(n_frames, 2048, device=device, dtype=dtype)
+    use_graph = device.type == 'cuda' and not args.no_graphs
+    if use_graph:
+        static_in = torch.empty(fb, 224, 224, 3, dtype=torch.uint8,
+                                device=device)
+        s = torch.cuda.Stream()
+        s.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(s), torch.no_grad():
+            for _ in range(2):
+                fwd(static_in)
+        torch.cuda.current_stream().wait_stream(s)
+        graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(graph), torch.no_grad():
+            static_out = fwd(static_in)
+
+    def step():
+        for st in range(0, n_frames, fb):
+            chunk = host[st:st + fb]
+            if use_graph and chunk.shape[0] == fb:
+                static_in.copy_(chunk, non_blocking=True)
+                graph.replay()
+                feats[st:st + fb].copy_(static_out)
+            else:
+                feats[st:st + chunk.shape[0]].copy_(
+                    fwd(chunk.to(device, non_blocking=True)))
+        return feats.float().cpu()
+
+    with torch.no_grad():
+        for _ in range(args.warmup):
+            step()
+        sync(device, world)
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step()
+        sync(device, world)
+        dt = time.perf_counter() - t0
+    dt = max_over_ranks(dt, device, world)
+    total = n_frames * world * args.steps
+    return {
+        'metric': 'frames/sec ResNet-50 fix_2',
+        'value': total / dt,
+        'unit': 'frames/sec',
+        'ms_per_step': dt / args.steps * 1000.0,
+        'config': {'model': 'ResNet-50', 'global_batch': n_frames * world,
+                   'seq_len': 1, 'resolution': 224,
+                   'parallelism': f'dp{world}'},
+    }
+
+
+# -------------------------------------------------- VGGish + R(2+1)D dual
+def bench_vggish_r21d(args, device, dtype, rank, world):
+    """BASELINE.json config 5: dual-stream audio (VGGish) + visual
+    (R(2+1)D-18 on 16-frame stacks) clips/sec."""
+    from video_features_amd.models.vggish import VGGish
+    from video_features_amd.models.r21d import R2Plus1D18
+    from video_features_amd import transforms as T
+    torch.manual_seed(0)
+    vgg = VGGish().to(device, dtype).eval()
+    r21d = R2Plus1D18().to(device, dtype).eval()
+    if world > 1:
+        from video_features_amd.runtime.dist import broadcast_models
+        broadcast_models({'a': vgg, 'b': r21d})
+
+    clips = max(args.clips_per_step, 1)
+    g = torch.Generator().manual_seed(rank + 1)
+    # visual: 16-frame 112x112 stacks; audio: 0.96 s log-mel examples
+    host_v = torch.rand(clips, 3, 16, 112, 112, generator=g)
+    host_a = torch.rand(clips, 96, 64, generator=g)
+    vis = host_v.to(device, dtype)
+    aud = host_a.to(device, dtype)
+
+    def step():
+        fa = vgg(aud)
+        fv = r21d.forward_features(vis)
+        return torch.cat([fv, fa.to(fv.dtype)], dim=1).float().cpu()
+
+    with torch.no_grad():
+        for _ in range(args.warmup):
+            step()
+        sync(device, world)
+        t0 = time.perf_counter()
+        for _ in range(args.steps):
+            step()
+        sync(device, world)
+        dt = time.perf_counter() - t0
+    dt = max_over_ranks(dt, device, world)
+    total = clips * world * args.steps
+    return {
+        'metric': 'clips/sec VGGish+R(2+1)D dual-stream',
+        'value': total / dt,
+        'unit': 'clips/sec',
+        'ms_per_step': dt / args.steps * 1000.0,
+        'config': {'model': 'VGGish+R(2+1)D-18', 'global_batch': clips * world,
+                   'seq_len': 16, 'resolution': 112,
+                   'parallelism': f'dp{world}'},
+    }
+
+
 def main():
     p = argparse.ArgumentParser()
     p.add_argument('--gpus', type=int, default=1)
     p.add_argument('--steps', type=int, default=16)
     p.add_argument('--warmup', type=int, default=4)
-    p.add_argument('--model', choices=['clip', 'i3d_raft'], default='clip')
+    p.add_argument('--model', choices=['clip', 'i3d_raft', 'resnet50', 'vggish_r21d'],
+                   default='clip')
     p.add_argument('--videos-per-step', type=int, default=32,
                    help='CLIP: synthetic videos (x12 frames) per rank per step')
     p.add_argument('--frame-batch', type=int, default=192,
@@ -280,10 +409,9 @@ def main():
         args.raft_iters = min(args.raft_iters, 3)
     setup(world, rank, local_rank, device)
 
-    if args.model == 'clip':
-        res = bench_clip(args, device, dtype, rank, world)
-    else:
-        res = bench_i3d_raft(args, device, dtype, rank, world)
+    bench_fns = {'clip': bench_clip, 'i3d_raft': bench_i3d_raft,
+                 'resnet50': bench_resnet, 'vggish_r21d': bench_vggish_r21d}
+    res = bench_fns[args.model](args, device, dtype, rank, world)
 
     if rank == 0:
         out = {
