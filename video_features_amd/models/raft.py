@@ -252,6 +252,11 @@ class RAFT(nn.Module):
                          memory_format=mf)
         rhx = torch.empty_like(hx, memory_format=mf)
         hx[:, :self.hdim] = torch.tanh(net)
+        # context features are loop-invariant: written once into both
+        # buffers; only the motion half is refreshed per iteration
+        ctx_end = self.hdim + self.cdim
+        hx[:, self.hdim:ctx_end] = inp
+        rhx[:, self.hdim:ctx_end] = inp
 
         coords0 = self.coords_grid(b, h8, w8, image1.device, torch.float32)
         coords1 = coords0.clone()
@@ -261,9 +266,8 @@ class RAFT(nn.Module):
             corr = corr_fn(coords1, nhwc=self.nhwc, out_dtype=dtype)
             flow = (coords1 - coords0).to(dtype).contiguous(memory_format=mf)
             motion = ub.encoder(flow, corr)
-            x = torch.cat([inp, motion], dim=1)
-            hx[:, self.hdim:] = x
-            rhx[:, self.hdim:] = x
+            hx[:, ctx_end:] = motion
+            rhx[:, ctx_end:] = motion
             ub.gru(hx, rhx, self.nhwc)
             net = hx[:, :self.hdim]
             delta = ub.flow_head(net)
