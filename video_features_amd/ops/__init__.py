@@ -214,7 +214,8 @@ def corr_lookup(pyramid, coords: torch.Tensor, radius: int = 4,
         out_dtype = coords.dtype
     if _use_hip(coords) and radius == 4:
         return _ext.corr_lookup(list(pyramid), coords.contiguous(), nhwc,
-                                out_dtype)
+                                out_dtype,
+                                bool(os.environ.get('VFA_CORR_GMEM')))
     # torch reference: per-level grid_sample of the displacement window
     b, _, h, w = coords.shape
     r = radius

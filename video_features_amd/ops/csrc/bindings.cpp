@@ -207,7 +207,7 @@ torch::Tensor mfma_gemm16(torch::Tensor a, torch::Tensor b) {
 
 torch::Tensor corr_lookup(std::vector<torch::Tensor> pyramid,
                           torch::Tensor coords, bool nhwc,
-                          torch::ScalarType out_dtype) {
+                          torch::ScalarType out_dtype, bool force_gmem) {
   // pyramid: up to 4 levels of (N, 1, h_l, w_l) fp32, N = B*H*W; coords
   // (B, 2, H, W) fp32 pixel units at level 0.  Returns (B, L*81, H, W)
   // (contiguous, or channels_last when nhwc) in out_dtype.
@@ -246,8 +246,8 @@ torch::Tensor corr_lookup(std::vector<torch::Tensor> pyramid,
   vfa_corr_lookup(ptrs[0], ptrs[1], ptrs[2], ptrs[3], cc.data_ptr(),
                   out.data_ptr(), levels, hs[0], ws[0], hs[1], ws[1], hs[2],
                   ws[2], hs[3], ws[3], npix, h * w,
-                  (int)(lds_floats <= 16384 ? lds_floats : 0), nhwc ? 1 : 0,
-                  tag, current_stream());
+                  (int)((lds_floats <= 16384 && !force_gmem) ? lds_floats : 0),
+                  nhwc ? 1 : 0, tag, current_stream());
   return out;
 }
 
