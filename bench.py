@@ -180,46 +180,48 @@ def bench_i3d_raft(args, device, dtype, rank, world):
 
     gathered = [None]
 
-    def clip_fwd(frames_u8_dev):
-        x = frames_u8_dev.permute(0, 3, 1, 2).to(dtype)   # (65, 3, 224, 224)
-        flow = raft(x[:-1], x[1:], test_mode=True)
-        rgb_in = T.scale_to_pm1(x[:-1]).transpose(0, 1)[None]
-        flow_in = T.i3d_flow_preprocess(flow, 224).transpose(0, 1)[None]
+    def clips_fwd(frames_u8_dev):
+        # (clips, 65, 224, 224, 3) u8 — ALL clips batched through one RAFT
+        # pass (clips*64 frame pairs) and one I3D pass per stream (batch =
+        # clips): bigger GEMMs, 1/clips the kernel launches
+        x = frames_u8_dev.permute(0, 1, 4, 2, 3).to(dtype)
+        i1 = x[:, :-1].reshape(-1, 3, 224, 224)
+        i2 = x[:, 1:].reshape(-1, 3, 224, 224)
+        flow = raft(i1, i2, test_mode=True)          # (clips*64, 2, H, W)
+        rgb_in = T.scale_to_pm1(x[:, :-1]).transpose(1, 2)
+        flow_in = T.i3d_flow_preprocess(flow, 224) \
+            .reshape(clips, stack, 2, 224, 224).transpose(1, 2)
         if cl3d:
             rgb_in = rgb_in.contiguous(memory_format=torch.channels_last_3d)
             flow_in = flow_in.contiguous(memory_format=torch.channels_last_3d)
         f_rgb = i3d_rgb.forward_features(rgb_in)
         f_flow = i3d_flow.forward_features(flow_in)
-        return torch.cat([f_rgb, f_flow], dim=1)
+        return torch.cat([f_rgb, f_flow], dim=1)     # (clips, 2048)
 
     use_graph = device.type == 'cuda' and not args.no_graphs
     if use_graph:
-        # hipGraph-capture the whole per-clip pipeline (RAFT 20-iteration
-        # loop + both I3D streams): replay removes thousands of per-clip
+        # hipGraph-capture the whole per-step pipeline (RAFT 20-iteration
+        # loop + both I3D streams): replay removes thousands of per-step
         # kernel-launch gaps
-        static_in = torch.empty(stack + 1, 224, 224, 3, dtype=torch.uint8,
-                                device=device)
+        static_in = torch.empty(clips, stack + 1, 224, 224, 3,
+                                dtype=torch.uint8, device=device)
         s = torch.cuda.Stream()
         s.wait_stream(torch.cuda.current_stream())
         with torch.cuda.stream(s), torch.no_grad():
             for _ in range(2):
-                clip_fwd(static_in)
+                clips_fwd(static_in)
         torch.cuda.current_stream().wait_stream(s)
         graph = torch.cuda.CUDAGraph()
         with torch.cuda.graph(graph), torch.no_grad():
-            static_out = clip_fwd(static_in)
+            static_out = clips_fwd(static_in)
 
     def step():
-        feats = []
-        for ci in range(clips):
-            if use_graph:
-                static_in.copy_(host[ci], non_blocking=True)
-                graph.replay()
-                feats.append(static_out.clone())
-            else:
-                frames = host[ci].to(device, non_blocking=True)
-                feats.append(clip_fwd(frames))
-        out = torch.cat(feats)
+        if use_graph:
+            static_in.copy_(host, non_blocking=True)
+            graph.replay()
+            out = static_out.clone()
+        else:
+            out = clips_fwd(host.to(device, non_blocking=True))
         # RCCL all-gather of stack features (BASELINE.json config 4)
         if world > 1:
             import torch.distributed as dist
