@@ -48,7 +48,10 @@ class ExtractI3D(BaseExtractor):
         if 'flow' in self.streams:
             models['flow'] = I3D(modality='flow').to(device, dtype).eval()
             if self.flow_type == 'raft':
-                models['flow_xtr'] = RAFT().to(device, dtype).eval()
+                raft = RAFT().to(device, dtype).eval()
+                if device.type == 'cuda':
+                    raft = raft.use_channels_last()
+                models['flow_xtr'] = raft
             elif self.flow_type == 'pwc':
                 models['flow_xtr'] = PWCNet().to(device, dtype).eval()
         return models
@@ -69,13 +72,16 @@ class ExtractI3D(BaseExtractor):
         x = T.resize_improved(x, RESIZE_SIDE, smaller_edge=True)
         return x, reader.fps
 
-    def _compute_flow(self, models, stack: torch.Tensor) -> torch.Tensor:
-        """(S+1, 3, H, W) frames → (S, 2, H, W) flow via RAFT or PWC."""
+    def _compute_flow(self, models, stacks: torch.Tensor) -> torch.Tensor:
+        """(B, S+1, 3, H, W) frames → (B*S, 2, H, W) flow via RAFT or PWC
+        (all frame pairs of the batch in one flow-net forward)."""
+        im1 = stacks[:, :-1].reshape(-1, *stacks.shape[2:])
+        im2 = stacks[:, 1:].reshape(-1, *stacks.shape[2:])
         if self.flow_type == 'raft':
-            padder = InputPadder(stack.shape)
-            im1, im2 = padder.pad(stack[:-1], stack[1:])
-            return padder.unpad(models['flow_xtr'](im1, im2, test_mode=True))
-        return models['flow_xtr'](stack[:-1], stack[1:])
+            padder = InputPadder(im1.shape)
+            p1, p2 = padder.pad(im1, im2)
+            return padder.unpad(models['flow_xtr'](p1, p2, test_mode=True))
+        return models['flow_xtr'](im1, im2)
 
     def _read_precomputed_flow(self, flow_dir: str, device, dtype,
                                count: int) -> torch.Tensor:
@@ -104,30 +110,39 @@ class ExtractI3D(BaseExtractor):
         if precomputed and 'flow' in self.streams:
             flow_all = self._read_precomputed_flow(video_path[1], device,
                                                    dtype, n - 1)
-        start = 0
-        while start + self.stack_size + 1 <= n:
-            stack = frames[start:start + self.stack_size + 1]
-            stack = stack.to(device=device, dtype=dtype, non_blocking=True)
+        ssz, step = self.stack_size, self.step_size
+        starts = [s for s in range(0, max(n - ssz, 1), step)
+                  if s + ssz + 1 <= n]
+        bs = max(1, self.cfg.batch_size or 1)
+        for i in range(0, len(starts), bs):
+            grp = starts[i:i + bs]
+            # (B, S+1, 3, H, W): batch of sliding windows — one flow-net
+            # forward over all pairs, one I3D forward per stream
+            stacks = torch.stack([frames[s:s + ssz + 1] for s in grp])
+            stacks = stacks.to(device=device, dtype=dtype, non_blocking=True)
+            b = stacks.shape[0]
             for stream in self.streams:
                 if stream == 'rgb':
-                    x = T.center_crop(stack[:-1], CROP)
+                    x = T.center_crop(
+                        stacks[:, :-1].reshape(-1, *stacks.shape[2:]), CROP)
                     x = T.scale_to_pm1(x)
                 else:
                     if flow_all is not None:
-                        flow = flow_all[start:start + self.stack_size]
+                        flow = torch.cat([flow_all[s:s + ssz] for s in grp])
                     else:
-                        flow = self._compute_flow(models, stack)
+                        flow = self._compute_flow(models, stacks)
                     x = T.i3d_flow_preprocess(flow, CROP)
-                clip = x.transpose(0, 1)[None]       # (1, C, T, H, W)
+                clip = x.reshape(b, ssz, *x.shape[1:]).transpose(1, 2)
                 f = models[stream].forward_features(clip)
                 feats[stream].append(f.float().cpu())
                 if self.show_pred:
                     from ..utils.labels import show_predictions_on_dataset
-                    print(f'{stream} stack @ {start}:')
-                    show_predictions_on_dataset(models[stream](clip).float().cpu(),
-                                                'kinetics')
-            ts.append(start / fps * 1000.0)
-            start += self.step_size
+                    logits = models[stream](clip).float().cpu()
+                    for j, s0 in enumerate(grp):
+                        print(f'{stream} stack @ {s0}:')
+                        show_predictions_on_dataset(logits[j:j + 1],
+                                                    'kinetics')
+            ts.extend(s / fps * 1000.0 for s in grp)
         out: Dict[str, np.ndarray] = {
             s: (torch.cat(feats[s]).numpy() if feats[s]
                 else np.zeros((0, I3D.FEAT_DIM), np.float32))

@@ -20,6 +20,8 @@ import torch
 import torch.nn.functional as F
 from torch import nn
 
+from .. import ops
+
 
 def _same_pad_1d(n: int, k: int, s: int) -> Tuple[int, int]:
     total = max(k - s, 0) if n % s == 0 else max(k - (n % s), 0)
@@ -78,8 +80,9 @@ class MaxPool3dSame(nn.Module):
         self.stride = _triple(stride)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return F.max_pool3d(tf_same_pad_3d(x, self.kernel, self.stride),
-                            self.kernel, self.stride)
+        # GPU: fused TF-SAME pool (no padded copy, no argmax indices);
+        # CPU fallback inside the op is the F.pad + max_pool3d reference
+        return ops.maxpool3d_same(x, self.kernel, self.stride)
 
 
 class Mixed(nn.Module):

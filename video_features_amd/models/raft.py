@@ -34,31 +34,55 @@ from .. import ops
 
 
 # --------------------------------------------------------------- encoders
+class FusedInstanceNorm2d(nn.Module):
+    """InstanceNorm2d(affine=False) with optional fused ReLU — one HIP
+    kernel on GPU via ``ops.instance_norm`` (torch lowers InstanceNorm to
+    batch-norm stats + transform + separate relu).  Stateless, so it is
+    checkpoint-compatible with ``nn.InstanceNorm2d``."""
+
+    def __init__(self, relu: bool = False):
+        super().__init__()
+        self.fuses_relu = relu
+
+    def forward(self, x):
+        if x.is_contiguous():
+            return ops.instance_norm(x, relu=self.fuses_relu, nhwc=False)
+        if x.is_contiguous(memory_format=torch.channels_last):
+            return ops.instance_norm(x, relu=self.fuses_relu, nhwc=True)
+        return ops.instance_norm(x.contiguous(), relu=self.fuses_relu,
+                                 nhwc=False)
+
+
+def _make_norm(norm: str, c: int, relu_after: bool = False):
+    if norm == 'instance':
+        return FusedInstanceNorm2d(relu=relu_after)
+    if norm == 'batch':
+        return nn.BatchNorm2d(c)
+    return nn.Identity()
+
+
+def _norm_act(norm_mod, relu, x):
+    y = norm_mod(x)
+    return y if getattr(norm_mod, 'fuses_relu', False) else relu(y)
+
+
 class ResidualBlock(nn.Module):
     def __init__(self, in_ch: int, out_ch: int, norm: str, stride: int = 1):
         super().__init__()
-
-        def make_norm(c):
-            if norm == 'instance':
-                return nn.InstanceNorm2d(c)
-            if norm == 'batch':
-                return nn.BatchNorm2d(c)
-            return nn.Identity()
-
         self.conv1 = nn.Conv2d(in_ch, out_ch, 3, stride, 1)
         self.conv2 = nn.Conv2d(out_ch, out_ch, 3, 1, 1)
-        self.norm1 = make_norm(out_ch)
-        self.norm2 = make_norm(out_ch)
+        self.norm1 = _make_norm(norm, out_ch, relu_after=True)
+        self.norm2 = _make_norm(norm, out_ch, relu_after=True)
         self.relu = nn.ReLU(inplace=True)
         if stride == 1 and in_ch == out_ch:
             self.downsample = None
         else:
             self.downsample = nn.Sequential(
-                nn.Conv2d(in_ch, out_ch, 1, stride), make_norm(out_ch))
+                nn.Conv2d(in_ch, out_ch, 1, stride), _make_norm(norm, out_ch))
 
     def forward(self, x):
-        y = self.relu(self.norm1(self.conv1(x)))
-        y = self.relu(self.norm2(self.conv2(y)))
+        y = _norm_act(self.norm1, self.relu, self.conv1(x))
+        y = _norm_act(self.norm2, self.relu, self.conv2(y))
         identity = x if self.downsample is None else self.downsample(x)
         return self.relu(identity + y)
 
@@ -68,12 +92,7 @@ class BasicEncoder(nn.Module):
 
     def __init__(self, output_dim: int = 256, norm: str = 'instance'):
         super().__init__()
-        if norm == 'instance':
-            self.norm1 = nn.InstanceNorm2d(64)
-        elif norm == 'batch':
-            self.norm1 = nn.BatchNorm2d(64)
-        else:
-            self.norm1 = nn.Identity()
+        self.norm1 = _make_norm(norm, 64, relu_after=True)
         self.conv1 = nn.Conv2d(3, 64, 7, 2, 3)
         self.relu1 = nn.ReLU(inplace=True)
         self.layer1 = nn.Sequential(ResidualBlock(64, 64, norm),
@@ -85,7 +104,7 @@ class BasicEncoder(nn.Module):
         self.conv2 = nn.Conv2d(128, output_dim, 1)
 
     def forward(self, x):
-        x = self.relu1(self.norm1(self.conv1(x)))
+        x = _norm_act(self.norm1, self.relu1, self.conv1(x))
         x = self.layer3(self.layer2(self.layer1(x)))
         return self.conv2(x)
 

@@ -281,6 +281,38 @@ def gru_out(q: torch.Tensor, z: torch.Tensor, hx: torch.Tensor,
     hx[:, :c] = (1 - z) * hx[:, :c] + z * torch.tanh(q)
 
 
+def instance_norm(x: torch.Tensor, eps: float = 1e-5, relu: bool = False,
+                  nhwc: bool = False) -> torch.Tensor:
+    """InstanceNorm2d (affine=False) with optional fused ReLU — ONE kernel
+    on GPU (torch lowers this to batch-norm stats + transform + separate
+    relu, three HBM round-trips)."""
+    if _use_hip(x):
+        return _ext.instance_norm2d(x, eps, relu, nhwc)
+    y = torch.nn.functional.instance_norm(x, eps=eps)
+    return torch.nn.functional.relu(y) if relu else y
+
+
+def maxpool3d_same(x: torch.Tensor, kernel, stride) -> torch.Tensor:
+    """TF-SAME max_pool3d (zero padding, asymmetric, extra cell at the end —
+    reference i3d_net.py:108-120).  GPU: one fused kernel, no padded copy,
+    no argmax indices."""
+    def _pad1(n, k, s):
+        total = max(k - s, 0) if n % s == 0 else max(k - (n % s), 0)
+        return total // 2, total - total // 2
+
+    t, h, w = x.shape[-3:]
+    pt, ph, pw = _pad1(t, kernel[0], stride[0]), _pad1(h, kernel[1], stride[1]),         _pad1(w, kernel[2], stride[2])
+    if _use_hip(x):
+        out_sz = [(t + pt[0] + pt[1] - kernel[0]) // stride[0] + 1,
+                  (h + ph[0] + ph[1] - kernel[1]) // stride[1] + 1,
+                  (w + pw[0] + pw[1] - kernel[2]) // stride[2] + 1]
+        return _ext.maxpool3d_same(x.contiguous(), list(kernel), list(stride),
+                                   [pt[0], ph[0], pw[0]], out_sz)
+    xp = torch.nn.functional.pad(
+        x, (pw[0], pw[1], ph[0], ph[1], pt[0], pt[1]))
+    return torch.nn.functional.max_pool3d(xp, tuple(kernel), tuple(stride))
+
+
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:
     """RAFT-style bilinear lookup: ``coords`` (B, Ho, Wo, 2) in *pixel* units,
     zero padding outside (reference models/raft/raft_src/utils/utils.py:57-71)."""

@@ -36,6 +36,11 @@ void vfa_gru_zr(const void*, const void*, void*, void*, int, int, int,
                 long long, int, int, hipStream_t);
 void vfa_gru_out(const void*, const void*, void*, int, int, int, long long,
                  int, int, hipStream_t);
+void vfa_instance_norm2d(const void*, void*, int, int, int, float, int, int,
+                         int, hipStream_t);
+void vfa_maxpool3d_same(const void*, void*, long long, int, int, int, int,
+                        int, int, int, int, int, int, int, int, int, int,
+                        int, int, hipStream_t);
 }
 
 namespace {
@@ -307,6 +312,40 @@ void gru_out(torch::Tensor q, torch::Tensor z, torch::Tensor hx, bool nhwc) {
               nhwc ? 1 : 0, dtype_tag(q), current_stream());
 }
 
+torch::Tensor instance_norm2d(torch::Tensor x, double eps, bool relu,
+                              bool nhwc) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+  if (nhwc) {
+    TORCH_CHECK(cl_contig(x), "channels_last expected");
+  } else {
+    TORCH_CHECK(x.is_contiguous());
+  }
+  auto out = torch::empty_like(x);
+  vfa_instance_norm2d(x.data_ptr(), out.data_ptr(), (int)x.size(0),
+                      (int)x.size(1), (int)(x.size(2) * x.size(3)),
+                      (float)eps, relu ? 1 : 0, nhwc ? 1 : 0, dtype_tag(x),
+                      current_stream());
+  return out;
+}
+
+torch::Tensor maxpool3d_same(torch::Tensor x, std::vector<int64_t> kernel,
+                             std::vector<int64_t> stride,
+                             std::vector<int64_t> pad_front,
+                             std::vector<int64_t> out_sz) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 5);
+  const long long bc = x.size(0) * x.size(1);
+  auto out = torch::empty({x.size(0), x.size(1), out_sz[0], out_sz[1],
+                           out_sz[2]}, x.options());
+  vfa_maxpool3d_same(x.data_ptr(), out.data_ptr(), bc, (int)x.size(2),
+                     (int)x.size(3), (int)x.size(4), (int)out_sz[0],
+                     (int)out_sz[1], (int)out_sz[2], (int)kernel[0],
+                     (int)kernel[1], (int)kernel[2], (int)stride[0],
+                     (int)stride[1], (int)stride[2], (int)pad_front[0],
+                     (int)pad_front[1], (int)pad_front[2], dtype_tag(x),
+                     current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -325,5 +364,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("convex_upsample", &convex_upsample);
   m.def("gru_zr", &gru_zr);
   m.def("gru_out", &gru_out);
+  m.def("instance_norm2d", &instance_norm2d);
+  m.def("maxpool3d_same", &maxpool3d_same);
   m.attr("gfx_arch") = "gfx950";
 }
