@@ -49,7 +49,9 @@ void launch_act(const void* in, void* out, long long n, hipStream_t stream) {
   constexpr int VEC = 16 / sizeof(T);
   long long nvec = (n + VEC - 1) / VEC;
   int block = 256;
-  int grid = (int)min((nvec + block - 1) / block, (long long)2048);
+  // one sweep, one iteration per thread where possible: the grid-stride
+  // loop's serialized load->store chain leaves HBM underused at this size
+  int grid = (int)min((nvec + block - 1) / block, (long long)65536);
   hipLaunchKernelGGL((act_kernel<T, VEC, KIND>), dim3(grid), dim3(block), 0,
                      stream, (const T*)in, (T*)out, n);
 }
