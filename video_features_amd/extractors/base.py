@@ -34,6 +34,8 @@ from ..runtime.progress import make_progress
 class BaseExtractor(torch.nn.Module):
     feature_type: str = ''
 
+    prof = None
+
     def __init__(self, args: Any, external_call: bool = False):
         super().__init__()
         cfg = Config.coerce(args)
@@ -61,6 +63,12 @@ class BaseExtractor(torch.nn.Module):
     def extract(self, device: torch.device, models: Any,
                 video_path) -> Dict[str, np.ndarray]:
         raise NotImplementedError
+
+    def _prof(self, stage: str):
+        from ..runtime.profiler import StageProfiler
+        if self.prof is None:
+            self.prof = StageProfiler(False)
+        return self.prof(stage)
 
     # ------------------------------------------------------------ helpers
     def compute_dtype(self, device: torch.device) -> torch.dtype:
@@ -100,6 +108,9 @@ class BaseExtractor(torch.nn.Module):
     @torch.no_grad()
     def forward(self, indices: torch.LongTensor) -> List[Dict[str, np.ndarray]]:
         device = indices.device
+        from ..runtime.profiler import StageProfiler
+        self.prof = StageProfiler(bool(getattr(self.cfg, 'profile', False)),
+                                  device)
         models = self.models_for(device)
         feats_list: List[Dict[str, np.ndarray]] = []
         for idx in indices.tolist():
@@ -113,9 +124,11 @@ class BaseExtractor(torch.nn.Module):
                 if self.external_call:
                     feats_list.append(feats_dict)
                 else:
-                    action_on_extraction(feats_dict, self._stem_path(video_path),
-                                         self.output_path, self.on_extraction,
-                                         self.output_direct, self.feature_type)
+                    with self.prof('sink'):
+                        action_on_extraction(
+                            feats_dict, self._stem_path(video_path),
+                            self.output_path, self.on_extraction,
+                            self.output_direct, self.feature_type)
             except KeyboardInterrupt:
                 raise
             except Exception as e:
@@ -124,6 +137,9 @@ class BaseExtractor(torch.nn.Module):
                       'Continuing extraction')
                 traceback.print_exc()
             self.progress.update()
+        rep = self.prof.report(f'{self.feature_type} shard on {device}')
+        if rep:
+            print(rep)
         return feats_list
 
     def _stem_path(self, video_path) -> str:

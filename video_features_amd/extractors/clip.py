@@ -37,18 +37,21 @@ class ExtractCLIP(BaseExtractor):
 
     def extract(self, device: torch.device, model,
                 video_path) -> Dict[str, np.ndarray]:
-        reader = open_video(video_path, self.tmp_path, self.extraction_fps)
-        fps = reader.fps
-        idxs = sample_indices(self.extract_method, reader.frame_count, fps)
-        frames_u8 = torch.from_numpy(reader.read_frames(idxs))
-        batch = T.clip_preprocess(frames_u8, model.cfg.input_resolution)
+        with self._prof('decode'):
+            reader = open_video(video_path, self.tmp_path, self.extraction_fps)
+            fps = reader.fps
+            idxs = sample_indices(self.extract_method, reader.frame_count, fps)
+            frames_u8 = torch.from_numpy(reader.read_frames(idxs))
+        with self._prof('preprocess'):
+            batch = T.clip_preprocess(frames_u8, model.cfg.input_resolution)
         dtype = self.compute_dtype(device)
         feats = []
-        for s in range(0, batch.shape[0], max(self.batch_size, 16)):
-            chunk = batch[s:s + max(self.batch_size, 16)]
-            chunk = chunk.to(device=device, dtype=dtype, non_blocking=True)
-            feats.append(model.encode_image(chunk).float().cpu())
-        features = torch.cat(feats).numpy()
+        with self._prof('infer'):
+            for s in range(0, batch.shape[0], max(self.batch_size, 16)):
+                chunk = batch[s:s + max(self.batch_size, 16)]
+                chunk = chunk.to(device=device, dtype=dtype, non_blocking=True)
+                feats.append(model.encode_image(chunk).float().cpu())
+            features = torch.cat(feats).numpy()
         return {
             self.feature_type: features,
             'fps': np.array(fps),

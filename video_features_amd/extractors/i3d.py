@@ -101,7 +101,8 @@ class ExtractI3D(BaseExtractor):
                 video_path) -> Dict[str, np.ndarray]:
         precomputed = isinstance(video_path, tuple)
         vid_path = video_path[0] if precomputed else video_path
-        frames, fps = self._read_resized_frames(vid_path)
+        with self._prof('decode'):
+            frames, fps = self._read_resized_frames(vid_path)
         dtype = self.compute_dtype(device)
         n = frames.shape[0]
         feats: Dict[str, List] = {s: [] for s in self.streams}
@@ -130,10 +131,12 @@ class ExtractI3D(BaseExtractor):
                     if flow_all is not None:
                         flow = torch.cat([flow_all[s:s + ssz] for s in grp])
                     else:
-                        flow = self._compute_flow(models, stacks)
+                        with self._prof('flow'):
+                            flow = self._compute_flow(models, stacks)
                     x = T.i3d_flow_preprocess(flow, CROP)
                 clip = x.reshape(b, ssz, *x.shape[1:]).transpose(1, 2)
-                f = models[stream].forward_features(clip)
+                with self._prof('infer'):
+                    f = models[stream].forward_features(clip)
                 feats[stream].append(f.float().cpu())
                 if self.show_pred:
                     from ..utils.labels import show_predictions_on_dataset
