@@ -313,6 +313,26 @@ def maxpool3d_same(x: torch.Tensor, kernel, stride) -> torch.Tensor:
     return torch.nn.functional.max_pool3d(xp, tuple(kernel), tuple(stride))
 
 
+def maxpool2d_same(x: torch.Tensor, kernel, stride,
+                   nhwc: bool = False) -> torch.Tensor:
+    """Spatial TF-SAME max_pool2d (zero padding, extra cell at the end) —
+    the spatial half of I3D's TF-SAME 3D pools in the flattened
+    (B*T, C, H, W) path; the temporal half is a shifted elementwise max."""
+    def _pad1(n, k, s):
+        total = max(k - s, 0) if n % s == 0 else max(k - (n % s), 0)
+        return total // 2, total - total // 2
+
+    h, w = x.shape[-2:]
+    ph, pw = _pad1(h, kernel[0], stride[0]), _pad1(w, kernel[1], stride[1])
+    if _use_hip(x):
+        out_sz = [(h + ph[0] + ph[1] - kernel[0]) // stride[0] + 1,
+                  (w + pw[0] + pw[1] - kernel[1]) // stride[1] + 1]
+        return _ext.maxpool2d_same(x, list(kernel), list(stride),
+                                   [ph[0], pw[0]], out_sz, nhwc)
+    xp = torch.nn.functional.pad(x, (pw[0], pw[1], ph[0], ph[1]))
+    return torch.nn.functional.max_pool2d(xp, tuple(kernel), tuple(stride))
+
+
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:
     """RAFT-style bilinear lookup: ``coords`` (B, Ho, Wo, 2) in *pixel* units,
     zero padding outside (reference models/raft/raft_src/utils/utils.py:57-71)."""

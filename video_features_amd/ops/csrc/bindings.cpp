@@ -41,6 +41,9 @@ void vfa_instance_norm2d(const void*, void*, int, int, int, float, int, int,
 void vfa_maxpool3d_same(const void*, void*, long long, int, int, int, int,
                         int, int, int, int, int, int, int, int, int, int,
                         int, int, hipStream_t);
+void vfa_maxpool2d_same(const void*, void*, long long, int, int, int, int,
+                        int, int, int, int, int, int, int, int, int,
+                        hipStream_t);
 }
 
 namespace {
@@ -346,6 +349,30 @@ torch::Tensor maxpool3d_same(torch::Tensor x, std::vector<int64_t> kernel,
   return out;
 }
 
+torch::Tensor maxpool2d_same(torch::Tensor x, std::vector<int64_t> kernel,
+                             std::vector<int64_t> stride,
+                             std::vector<int64_t> pad_front,
+                             std::vector<int64_t> out_sz, bool nhwc) {
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4);
+  if (nhwc) {
+    TORCH_CHECK(cl_contig(x), "channels_last expected");
+  } else {
+    TORCH_CHECK(x.is_contiguous());
+  }
+  auto out = nhwc
+      ? torch::empty({x.size(0), out_sz[0], out_sz[1], x.size(1)},
+                     x.options()).permute({0, 3, 1, 2})
+      : torch::empty({x.size(0), x.size(1), out_sz[0], out_sz[1]},
+                     x.options());
+  vfa_maxpool2d_same(x.data_ptr(), out.data_ptr(), x.size(0),
+                     (int)x.size(1), (int)x.size(2), (int)x.size(3),
+                     (int)out_sz[0], (int)out_sz[1], (int)kernel[0],
+                     (int)kernel[1], (int)stride[0], (int)stride[1],
+                     (int)pad_front[0], (int)pad_front[1], nhwc ? 1 : 0,
+                     dtype_tag(x), current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -366,5 +393,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gru_out", &gru_out);
   m.def("instance_norm2d", &instance_norm2d);
   m.def("maxpool3d_same", &maxpool3d_same);
+  m.def("maxpool2d_same", &maxpool2d_same);
   m.attr("gfx_arch") = "gfx950";
 }

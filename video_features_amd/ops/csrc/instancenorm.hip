@@ -99,13 +99,16 @@ template <typename T>
 void launch_in2d(const void* x, void* out, int b, int c, int hw, float eps,
                  int relu, int nhwc, hipStream_t stream) {
   if (nhwc) {
+    // 16 pixel sub-rows per block: enough waves in flight to hide the
+    // strided-load latency of the two passes (941us -> bandwidth-bound
+    // at ROWS=4 this kernel was 3x off the HBM floor)
     const dim3 grid((c + 63) / 64, b);
-    const dim3 block(64, 4);
+    const dim3 block(64, 16);
     if (relu)
-      hipLaunchKernelGGL((in2d_nhwc_kernel<T, true, 4>), grid, block, 0,
+      hipLaunchKernelGGL((in2d_nhwc_kernel<T, true, 16>), grid, block, 0,
                          stream, (const T*)x, (T*)out, c, hw, eps);
     else
-      hipLaunchKernelGGL((in2d_nhwc_kernel<T, false, 4>), grid, block, 0,
+      hipLaunchKernelGGL((in2d_nhwc_kernel<T, false, 16>), grid, block, 0,
                          stream, (const T*)x, (T*)out, c, hw, eps);
   } else {
     const dim3 grid(b * c);

@@ -50,9 +50,80 @@ __global__ void maxpool3d_same_kernel(const T* __restrict__ x,
   }
 }
 
+// spatial 2D TF-SAME max pool (zero padding), NCHW or NHWC — used by the
+// flattened (B*T, C, H, W) I3D path where the temporal max is a separate
+// shifted elementwise maximum
+template <typename T, bool NHWC>
+__global__ void maxpool2d_same_kernel(const T* __restrict__ x,
+                                      T* __restrict__ out, long long n,
+                                      int c, int ih, int iw, int oh, int ow,
+                                      int kh, int kw, int sh, int sw, int ph,
+                                      int pw) {
+  const long long ohwc = (long long)oh * ow * c;
+  const long long total = n * ohwc;
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    long long ni;
+    int yo, xo, ci;
+    if (NHWC) {
+      ni = i / ohwc;
+      const long long r = i % ohwc;
+      yo = (int)(r / ((long long)ow * c));
+      xo = (int)((r / c) % ow);
+      ci = (int)(r % c);
+    } else {
+      ni = i / ohwc;
+      const long long r = i % ohwc;
+      ci = (int)(r / ((long long)oh * ow));
+      yo = (int)((r / ow) % oh);
+      xo = (int)(r % ow);
+    }
+    const int y0 = yo * sh - ph, x0 = xo * sw - pw;
+    const T* xb = x + ni * (long long)c * ih * iw;
+    float m = -INFINITY;
+    bool padded = false;
+    for (int dy = 0; dy < kh; ++dy) {
+      const int y = y0 + dy;
+      if (y < 0 || y >= ih) { padded = true; continue; }
+      for (int dx = 0; dx < kw; ++dx) {
+        const int xx = x0 + dx;
+        if (xx < 0 || xx >= iw) { padded = true; continue; }
+        const long long xi = NHWC ? (((long long)y * iw + xx) * c + ci)
+                                  : (((long long)ci * ih + y) * iw + xx);
+        m = fmaxf(m, to_f32<T>(xb[xi]));
+      }
+    }
+    if (padded) m = fmaxf(m, 0.f);
+    out[i] = from_f32<T>(m);
+  }
+}
+
 }  // namespace
 
 extern "C" {
+
+void vfa_maxpool2d_same(const void* x, void* out, long long n, int c, int ih,
+                        int iw, int oh, int ow, int kh, int kw, int sh,
+                        int sw, int ph, int pw, int nhwc, int dtype,
+                        hipStream_t stream) {
+  const long long total = n * (long long)c * oh * ow;
+  const int grid = (int)min((total + 255) / 256, (long long)65536);
+#define VFA_MP2_CASE(T)                                                       \
+  if (nhwc)                                                                   \
+    hipLaunchKernelGGL((maxpool2d_same_kernel<T, true>), dim3(grid),          \
+                       dim3(256), 0, stream, (const T*)x, (T*)out, n, c, ih,  \
+                       iw, oh, ow, kh, kw, sh, sw, ph, pw);                   \
+  else                                                                        \
+    hipLaunchKernelGGL((maxpool2d_same_kernel<T, false>), dim3(grid),         \
+                       dim3(256), 0, stream, (const T*)x, (T*)out, n, c, ih,  \
+                       iw, oh, ow, kh, kw, sh, sw, ph, pw);
+  switch (dtype) {
+    case VFA_F32: VFA_MP2_CASE(float) break;
+    case VFA_BF16: VFA_MP2_CASE(__hip_bfloat16) break;
+    case VFA_F16: VFA_MP2_CASE(__half) break;
+  }
+#undef VFA_MP2_CASE
+}
 
 void vfa_maxpool3d_same(const void* x, void* out, long long bc, int it,
                         int ih, int iw, int ot, int oh, int ow, int kt,

@@ -30,7 +30,9 @@ namespace {
 // Pyramid: LV levels, level l is (N, 1, h[l], w[l]) fp32, N = B*H*W planes
 // (one correlation plane per query pixel).  coords: (B, 2, H, W) fp32 pixel
 // units at level 0.  out: (B, LV*81, H, W) or NHWC (B, H, W, LV*81) in T.
-template <typename T, int RAD, bool NHWC>
+// PPB query pixels per block (one 64-lane wave each): 16 waves in flight
+// per block hide the LDS-staging load latency
+template <typename T, int RAD, bool NHWC, int PPB>
 __global__ void corr_lookup_lds_kernel(const float* __restrict__ l0,
                                        const float* __restrict__ l1,
                                        const float* __restrict__ l2,
@@ -38,12 +40,13 @@ __global__ void corr_lookup_lds_kernel(const float* __restrict__ l0,
                                        const float* __restrict__ coords,
                                        T* __restrict__ out, int levels,
                                        int4 hs, int4 ws, long long npix,
-                                       int hw) {
-  extern __shared__ float lds[];
+                                       int hw, int lds_floats) {
+  extern __shared__ float lds_all[];
   const int K = 2 * RAD + 1;          // 9
   const int KK = K * K;               // 81 taps per level
-  const long long pix = blockIdx.x;   // one block per query pixel
+  const long long pix = blockIdx.x * (long long)PPB + threadIdx.y;
   if (pix >= npix) return;
+  float* lds = lds_all + (long long)threadIdx.y * lds_floats;
   const int tid = threadIdx.x;
 
   const int lh[4] = {hs.x, hs.y, hs.z, hs.w};
@@ -57,10 +60,10 @@ __global__ void corr_lookup_lds_kernel(const float* __restrict__ l0,
     const int n = lh[l] * lw[l];
     const float* src = lp[l] + pix * (long long)n;
     loff[l] = off;
-    for (int i = tid; i < n; i += blockDim.x) lds[off + i] = src[i];
+    for (int i = tid; i < n; i += 64) lds[off + i] = src[i];
     off += n;
   }
-  __syncthreads();
+  __builtin_amdgcn_wave_barrier();  // each wave stages its own pixel's planes
 
   const long long b = pix / hw;
   const int p = (int)(pix % hw);
@@ -68,7 +71,7 @@ __global__ void corr_lookup_lds_kernel(const float* __restrict__ l0,
   const float cy = coords[(b * 2 + 1) * hw + p];
 
   const int ctot = levels * KK;
-  for (int c = tid; c < ctot; c += blockDim.x) {
+  for (int c = tid; c < ctot; c += 64) {
     const int l = c / KK, t = c % KK;
     const int dy = t / K - RAD, dx = t % K - RAD;
     const float inv = 1.0f / (float)(1 << l);
@@ -269,16 +272,20 @@ void launch_corr_lookup(const float* l0, const float* l1, const float* l2,
                         const float* l3, const float* coords, void* out,
                         int levels, int4 hs, int4 ws, long long npix, int hw,
                         int lds_floats, int nhwc, hipStream_t stream) {
-  if (lds_floats > 0 && lds_floats * 4 <= 64 * 1024) {
-    const dim3 grid((unsigned)npix);
+  constexpr int PPB = 8;
+  if (lds_floats > 0 && lds_floats * 4 * PPB <= 64 * 1024) {
+    const dim3 grid((unsigned)((npix + PPB - 1) / PPB));
+    const dim3 block(64, PPB);
     if (nhwc)
-      hipLaunchKernelGGL((corr_lookup_lds_kernel<T, 4, true>), grid, dim3(64),
-                         lds_floats * 4, stream, l0, l1, l2, l3, coords,
-                         (T*)out, levels, hs, ws, npix, hw);
+      hipLaunchKernelGGL((corr_lookup_lds_kernel<T, 4, true, PPB>), grid,
+                         block, lds_floats * 4 * PPB, stream, l0, l1, l2, l3,
+                         coords, (T*)out, levels, hs, ws, npix, hw,
+                         lds_floats);
     else
-      hipLaunchKernelGGL((corr_lookup_lds_kernel<T, 4, false>), grid, dim3(64),
-                         lds_floats * 4, stream, l0, l1, l2, l3, coords,
-                         (T*)out, levels, hs, ws, npix, hw);
+      hipLaunchKernelGGL((corr_lookup_lds_kernel<T, 4, false, PPB>), grid,
+                         block, lds_floats * 4 * PPB, stream, l0, l1, l2, l3,
+                         coords, (T*)out, levels, hs, ws, npix, hw,
+                         lds_floats);
   } else {
     const long long total = npix * levels * 81;
     const int grid = (int)min((total + 255) / 256, (long long)16384);
