@@ -131,3 +131,48 @@ def test_vggish_sidecar_wav(tmp_path, frames16):
     ex = ExtractVGGish(cfg, external_call=True)
     out = ex(torch.arange(1))[0]
     assert out['vggish_torch'].shape == (1, 128)
+
+
+def test_i3d_flat_path_matches_5d_reference():
+    """The flattened-time backbone (conv2d decomposition + shifted temporal
+    add/max) must match the plain 5D conv3d/pool3d composition exactly."""
+    import torch.nn.functional as F
+    from video_features_amd.models.i3d import I3D
+
+    torch.manual_seed(3)
+    model = I3D(modality='rgb').eval()
+    # randomize BN stats so folding/normalization is exercised
+    for m in model.modules():
+        if isinstance(m, torch.nn.BatchNorm3d):
+            m.running_mean.normal_(0, 0.3)
+            m.running_var.uniform_(0.5, 2.0)
+    x = torch.randn(2, 3, 10, 64, 64)
+
+    def ref_backbone(m, x):
+        x = m.conv3d_1a_7x7(x)
+        x = m.maxPool3d_2a_3x3.forward(x)
+        x = m.conv3d_2c_3x3.forward(m.conv3d_2b_1x1.forward(x))
+        x = m.maxPool3d_3a_3x3.forward(x)
+
+        def mixed(mm, x):
+            return torch.cat([mm.b0.forward(x),
+                              mm.b1[1].forward(mm.b1[0].forward(x)),
+                              mm.b2[1].forward(mm.b2[0].forward(x)),
+                              mm.b3[1].forward(mm.b3[0].forward(x))], 1)
+        x = mixed(m.mixed_3c, mixed(m.mixed_3b, x))
+        x = m.maxPool3d_4a_3x3.forward(x)
+        for blk in (m.mixed_4b, m.mixed_4c, m.mixed_4d, m.mixed_4e,
+                    m.mixed_4f):
+            x = mixed(blk, x)
+        x = m.maxPool3d_5a_2x2.forward(x)
+        x = mixed(m.mixed_5c, mixed(m.mixed_5b, x))
+        kt = min(2, x.shape[2])
+        x = F.avg_pool3d(x, (kt, min(7, x.shape[3]), min(7, x.shape[4])))
+        return x.mean(dim=(2, 3, 4))
+
+    with torch.no_grad():
+        out = model.forward_features(x)
+        ref = ref_backbone(model, x)
+    assert out.shape == ref.shape == (2, 1024)
+    assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4), \
+        (out - ref).abs().max().item()

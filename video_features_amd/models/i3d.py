@@ -43,6 +43,83 @@ def _triple(v) -> Tuple[int, int, int]:
     return (v, v, v) if isinstance(v, int) else tuple(v)
 
 
+# ----------------------------------------------------- flattened-time path
+# On MI355X, MIOpen lowers NCDHW conv3d through Im3d2Col + GEMM (profiled in
+# profiles/) — an im2col materialization 27x the input for 3x3x3 kernels.
+# Instead the whole backbone below the stem runs on (B*T, C, H, W)
+# channels_last tensors: 1x1x1 convs become plain GEMM-shaped conv2d, a
+# 3x3x3 conv becomes ONE merged conv2d with 3*O outputs (the three temporal
+# taps) followed by a shifted add over t, and the TF-SAME 3D pools separate
+# exactly into a spatial 2D pool + a shifted temporal maximum (max over a
+# separable window commutes).
+
+def flatten_time(x: torch.Tensor) -> torch.Tensor:
+    """(B, C, T, H, W) → (B*T, C, H, W) channels_last, one copy."""
+    b, c, t, h, w = x.shape
+    y = x.permute(0, 2, 3, 4, 1).reshape(b * t, h, w, c)
+    return y.permute(0, 3, 1, 2)
+
+
+def unflatten_time(xf: torch.Tensor, b: int) -> torch.Tensor:
+    """(B*T, C, H, W) channels_last → (B, C, T, H, W) contiguous."""
+    bt, c, h, w = xf.shape
+    t = bt // b
+    return xf.permute(0, 2, 3, 1).reshape(b, t, h, w, c) \
+        .permute(0, 4, 1, 2, 3).contiguous()
+
+
+def _cl_empty(bt: int, c: int, h: int, w: int, like: torch.Tensor):
+    return torch.empty(bt, c, h, w, device=like.device, dtype=like.dtype,
+                       memory_format=torch.channels_last)
+
+
+def temporal_merge3(y: torch.Tensor, b: int) -> torch.Tensor:
+    """y (B*T, 3O, H, W): the three temporal-tap conv outputs stacked along
+    channels → (B*T, O, H, W) with out[t] = y0[t-1] + y1[t] + y2[t+1]
+    (zero temporal TF-SAME padding; bias was applied on the middle tap)."""
+    bt, c3, h, w = y.shape
+    o = c3 // 3
+    t = bt // b
+    out = _cl_empty(bt, o, h, w, y)
+    out.copy_(y[:, o:2 * o])
+    y5 = y.view(b, t, c3, h, w)
+    o5 = out.view(b, t, o, h, w)
+    if t > 1:
+        o5[:, 1:] += y5[:, :-1, :o]
+        o5[:, :-1] += y5[:, 1:, 2 * o:]
+    return out
+
+
+def temporal_max(xf: torch.Tensor, b: int, kt: int, st: int) -> torch.Tensor:
+    """Shifted maximum along t with TF-SAME zero padding:
+    (B*T, C, H, W) → (B*T', C, H, W), T' = ceil(T/st)."""
+    bt, c, h, w = xf.shape
+    t = bt // b
+    p0, p1 = _same_pad_1d(t, kt, st)
+    to = (t + p0 + p1 - kt) // st + 1
+    out = _cl_empty(b * to, c, h, w, xf)
+    out.fill_(float('-inf'))
+    x5 = xf.view(b, t, c, h, w)
+    o5 = out.view(b, to, c, h, w)
+    for dt in range(kt):
+        j_lo = max(0, -(-(p0 - dt) // st))
+        j_hi = min(to - 1, (t - 1 + p0 - dt) // st)
+        if j_lo > j_hi:
+            continue
+        s_lo = j_lo * st - p0 + dt
+        src = x5[:, s_lo:s_lo + (j_hi - j_lo) * st + 1:st]
+        dst = o5[:, j_lo:j_hi + 1]
+        torch.maximum(dst, src, out=dst)
+    # zero padding participates in the max at the temporal edges
+    head = -(-p0 // st)                       # windows missing leading taps
+    if head > 0:
+        o5[:, :head].clamp_(min=0)
+    j_tail = -(-(t + p0 - kt + 1) // st)      # windows past the last frame
+    if j_tail < to:
+        o5[:, j_tail:].clamp_(min=0)
+    return out
+
+
 class Unit3D(nn.Module):
     """Conv3d + BN + ReLU with TF-SAME padding
     (the reference's ``Unit3Dpy``, i3d_net.py:37-105)."""
@@ -72,6 +149,37 @@ class Unit3D(nn.Module):
             x = self.bn(x)
         return F.relu(x, inplace=True) if self.activation else x
 
+    def forward_flat(self, xf: torch.Tensor, b: int) -> torch.Tensor:
+        """Flattened-time forward on (B*T, C, H, W) channels_last (see the
+        module comment): 1x1x1 → conv2d GEMM; 3x3x3 → one merged conv2d
+        (3*O outputs) + temporal shift-add.  Only valid for the stride-1
+        units (everything but the stem)."""
+        kt = self.kernel[0]
+        w5 = self.conv.weight
+        bias = self.conv.bias
+        if kt == 1:
+            x = F.conv2d(xf, w5[:, :, 0], bias,
+                         padding=(self.kernel[1] // 2, self.kernel[2] // 2))
+        else:  # 3x3x3, stride 1
+            o = w5.shape[0]
+            wcat = w5.permute(2, 0, 1, 3, 4).reshape(
+                3 * o, w5.shape[1], self.kernel[1], self.kernel[2])
+            if bias is not None:
+                bcat = torch.zeros(3 * o, device=bias.device,
+                                   dtype=bias.dtype)
+                bcat[o:2 * o] = bias
+            else:
+                bcat = None
+            y = F.conv2d(xf, wcat, bcat,
+                         padding=(self.kernel[1] // 2, self.kernel[2] // 2))
+            x = temporal_merge3(y, b)
+        if isinstance(self.bn, nn.BatchNorm3d):
+            # BatchNorm3d == BatchNorm2d per channel on the flattened view
+            bn = self.bn
+            x = F.batch_norm(x, bn.running_mean, bn.running_var, bn.weight,
+                             bn.bias, bn.training, bn.momentum, bn.eps)
+        return F.relu(x, inplace=True) if self.activation else x
+
 
 class MaxPool3dSame(nn.Module):
     def __init__(self, kernel, stride):
@@ -83,6 +191,17 @@ class MaxPool3dSame(nn.Module):
         # GPU: fused TF-SAME pool (no padded copy, no argmax indices);
         # CPU fallback inside the op is the F.pad + max_pool3d reference
         return ops.maxpool3d_same(x, self.kernel, self.stride)
+
+    def forward_flat(self, xf: torch.Tensor, b: int) -> torch.Tensor:
+        # separable window: spatial TF-SAME 2D pool, then shifted temporal
+        # maximum (exact — max commutes over the separable window)
+        nhwc = xf.is_contiguous(memory_format=torch.channels_last) \
+            and not xf.is_contiguous()
+        y = ops.maxpool2d_same(xf, self.kernel[1:], self.stride[1:],
+                               nhwc=nhwc)
+        if self.kernel[0] > 1 or self.stride[0] > 1:
+            y = temporal_max(y, b, self.kernel[0], self.stride[0])
+        return y
 
 
 class Mixed(nn.Module):
@@ -102,6 +221,15 @@ class Mixed(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return torch.cat([self.b0(x), self.b1(x), self.b2(x), self.b3(x)], 1)
+
+    def forward_flat(self, xf: torch.Tensor, b: int) -> torch.Tensor:
+        outs = [self.b0.forward_flat(xf, b)]
+        for branch in (self.b1, self.b2, self.b3):
+            y = xf
+            for m in branch:
+                y = m.forward_flat(y, b)
+            outs.append(y)
+        return torch.cat(outs, 1)
 
 
 class I3D(nn.Module):
@@ -133,15 +261,21 @@ class I3D(nn.Module):
                                     activation=False, use_bias=True)
 
     def _backbone(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.conv3d_1a_7x7(x)
-        x = self.maxPool3d_2a_3x3(x)
-        x = self.conv3d_2c_3x3(self.conv3d_2b_1x1(x))
-        x = self.maxPool3d_3a_3x3(x)
-        x = self.mixed_3c(self.mixed_3b(x))
-        x = self.maxPool3d_4a_3x3(x)
-        x = self.mixed_4f(self.mixed_4e(self.mixed_4d(self.mixed_4c(self.mixed_4b(x)))))
-        x = self.maxPool3d_5a_2x2(x)
-        x = self.mixed_5c(self.mixed_5b(x))
+        b = x.shape[0]
+        x = self.conv3d_1a_7x7(x)              # stem stays a real conv3d
+        xf = flatten_time(x)                   # (B*T, C, H, W) channels_last
+        xf = self.maxPool3d_2a_3x3.forward_flat(xf, b)
+        xf = self.conv3d_2b_1x1.forward_flat(xf, b)
+        xf = self.conv3d_2c_3x3.forward_flat(xf, b)
+        xf = self.maxPool3d_3a_3x3.forward_flat(xf, b)
+        xf = self.mixed_3c.forward_flat(self.mixed_3b.forward_flat(xf, b), b)
+        xf = self.maxPool3d_4a_3x3.forward_flat(xf, b)
+        for m in (self.mixed_4b, self.mixed_4c, self.mixed_4d, self.mixed_4e,
+                  self.mixed_4f):
+            xf = m.forward_flat(xf, b)
+        xf = self.maxPool3d_5a_2x2.forward_flat(xf, b)
+        xf = self.mixed_5c.forward_flat(self.mixed_5b.forward_flat(xf, b), b)
+        x = unflatten_time(xf, b)
         # avg pool (2, 7, 7), stride 1 (reference i3d_net.py:229-235)
         kt = min(2, x.shape[2])
         return F.avg_pool3d(x, (kt, min(7, x.shape[3]), min(7, x.shape[4])))
