@@ -1,0 +1,74 @@
+"""GPU numerics for the fused MFMA linear kernel (ops.linear_act) vs the
+plain torch fp32 reference."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope='module')
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device('cuda:0')
+
+
+def _hip_loaded():
+    from video_features_amd import ops
+    assert ops.hip_available()
+    return ops
+
+
+def _ref(x, w, b, act):
+    y = torch.nn.functional.linear(x.float(), w.float(),
+                                   b.float() if b is not None else None)
+    if act == 'relu':
+        return y.relu()
+    if act == 'quick_gelu':
+        return y * torch.sigmoid(1.702 * y)
+    if act == 'gelu':
+        return torch.nn.functional.gelu(y, approximate='tanh')
+    return y
+
+
+@pytest.mark.parametrize('m,n,k', [(256, 128, 64), (512, 384, 192),
+                                   (1000, 768, 3072), (9600, 3072, 768),
+                                   (129, 256, 128)])
+@pytest.mark.parametrize('act', ['none', 'relu', 'quick_gelu', 'gelu'])
+def test_linear_act(dev, m, n, k, act):
+    ops = _hip_loaded()
+    torch.manual_seed(0)
+    x = (torch.randn(m, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    b = torch.randn(n, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, b, act)
+    assert out.shape == (m, n) and out.dtype == torch.bfloat16
+    ref = _ref(x, w, b, act)
+    err = (out.float() - ref).abs()
+    rel = err.max().item() / max(ref.abs().max().item(), 1e-6)
+    assert rel < 3e-2, rel
+    # tight mean check: accumulation is fp32, only I/O is bf16
+    assert (err.mean() / ref.abs().mean().clamp_min(1e-6)).item() < 5e-3
+
+
+def test_linear_act_no_bias(dev):
+    ops = _hip_loaded()
+    torch.manual_seed(1)
+    x = torch.randn(384, 256, device=dev).to(torch.bfloat16)
+    w = torch.randn(128, 256, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, None, 'none')
+    ref = _ref(x, w, None, 'none')
+    rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert rel < 3e-2, rel
+
+
+def test_linear_act_3d_input(dev):
+    ops = _hip_loaded()
+    torch.manual_seed(2)
+    x = torch.randn(4, 50, 768, device=dev).to(torch.bfloat16)
+    w = torch.randn(3072, 768, device=dev).to(torch.bfloat16)
+    b = torch.zeros(3072, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, b, 'quick_gelu')
+    assert out.shape == (4, 50, 3072)
+    ref = _ref(x.reshape(-1, 768), w, b, 'quick_gelu').reshape(4, 50, 3072)
+    rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert rel < 3e-2, rel

@@ -1,0 +1,62 @@
+#!/usr/bin/env python3
+"""Microbench: ops.linear_act (fused MFMA GEMM) vs torch F.linear(+act)
+on the CLIP / VGGish / RAFT shapes.  Run on a GPU box."""
+import time
+
+import torch
+
+from video_features_amd import ops
+
+assert torch.cuda.is_available()
+dev = torch.device('cuda:0')
+
+SHAPES = [
+    # (M, N, K, act, label)
+    (9600, 3072, 768, 'quick_gelu', 'CLIP fc1 (fb192)'),
+    (9600, 768, 3072, 'none', 'CLIP fc2'),
+    (9600, 2304, 768, 'none', 'CLIP qkv'),
+    (9600, 768, 768, 'none', 'CLIP proj'),
+    (38400, 3072, 768, 'quick_gelu', 'CLIP fc1 (fb768)'),
+    (4096, 4096, 12288, 'relu', 'VGGish fc1'),
+    (200704, 256, 1920, 'none', 'RAFT GRU-zr-as-GEMM'),
+]
+
+
+def timeit(fn, iters=50):
+    for _ in range(10):
+        fn()
+    torch.cuda.synchronize()
+    t0 = time.perf_counter()
+    for _ in range(iters):
+        fn()
+    torch.cuda.synchronize()
+    return (time.perf_counter() - t0) / iters
+
+
+for m, n, k, act, label in SHAPES:
+    torch.manual_seed(0)
+    x = (torch.randn(m, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    b = torch.randn(n, device=dev).to(torch.bfloat16)
+
+    def torch_path():
+        y = torch.nn.functional.linear(x, w, b)
+        if act == 'relu':
+            y = y.relu()
+        elif act == 'quick_gelu':
+            y = y * torch.sigmoid(1.702 * y)
+        return y
+
+    def ours():
+        return ops.linear_act(x, w, b, act)
+
+    flops = 2.0 * m * n * k
+    tt = timeit(torch_path)
+    to = timeit(ours)
+    # correctness spot-check
+    d = (ours().float() - torch_path().float()).abs().max().item()
+    ref = torch_path().float().abs().max().item()
+    print(f'{label:<24} M{m:>7} N{n:>5} K{k:>6} {act:<11} '
+          f'torch {tt * 1e6:7.1f}us ({flops / tt / 1e12:6.1f} TF) | '
+          f'ours {to * 1e6:7.1f}us ({flops / to / 1e12:6.1f} TF) | '
+          f'x{tt / to:4.2f} relerr {d / ref:.2e}', flush=True)

@@ -61,7 +61,9 @@ class MLP(nn.Module):
         self.c_proj = nn.Linear(4 * width, width)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
-        return self.c_proj(ops.quick_gelu(self.c_fc(x)))
+        # fc1 + QuickGELU fused into one MFMA GEMM epilogue on GPU
+        h = ops.linear_act(x, self.c_fc.weight, self.c_fc.bias, 'quick_gelu')
+        return ops.linear_act(h, self.c_proj.weight, self.c_proj.bias)
 
 
 class ResidualAttentionBlock(nn.Module):

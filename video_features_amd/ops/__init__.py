@@ -333,6 +333,34 @@ def maxpool2d_same(x: torch.Tensor, kernel, stride,
     return torch.nn.functional.max_pool2d(xp, tuple(kernel), tuple(stride))
 
 
+_ACT_IDS = {'none': 0, 'relu': 1, 'quick_gelu': 2, 'gelu': 3}
+
+
+def linear_act(x: torch.Tensor, weight: torch.Tensor,
+               bias: Optional[torch.Tensor] = None,
+               act: str = 'none') -> torch.Tensor:
+    """``act(x @ weight^T + bias)`` — one fused MFMA GEMM on GPU (bf16,
+    128x128 tiles, glds-staged LDS with st_16x32 swizzle, bias+activation
+    in the epilogue).  Falls back to F.linear + activation on CPU or for
+    shapes outside K%64==0 / N%128==0."""
+    k = x.shape[-1]
+    n = weight.shape[0]
+    if (_use_hip(x) and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16 and k % 64 == 0
+            and n % 128 == 0):
+        x2 = x.reshape(-1, k).contiguous()
+        out = _ext.linear_act(x2, weight.contiguous(), bias, _ACT_IDS[act])
+        return out.reshape(*x.shape[:-1], n)
+    y = torch.nn.functional.linear(x, weight, bias)
+    if act == 'relu':
+        return torch.nn.functional.relu(y)
+    if act == 'quick_gelu':
+        return y * torch.sigmoid(1.702 * y)
+    if act == 'gelu':
+        return torch.nn.functional.gelu(y, approximate='tanh')
+    return y
+
+
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:
     """RAFT-style bilinear lookup: ``coords`` (B, Ho, Wo, 2) in *pixel* units,
     zero padding outside (reference models/raft/raft_src/utils/utils.py:57-71)."""

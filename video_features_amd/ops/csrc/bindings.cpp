@@ -44,6 +44,8 @@ void vfa_maxpool3d_same(const void*, void*, long long, int, int, int, int,
 void vfa_maxpool2d_same(const void*, void*, long long, int, int, int, int,
                         int, int, int, int, int, int, int, int, int,
                         hipStream_t);
+void vfa_linear_act(const void*, const void*, const void*, void*, int, int,
+                    int, int, hipStream_t);
 }
 
 namespace {
@@ -373,6 +375,29 @@ torch::Tensor maxpool2d_same(torch::Tensor x, std::vector<int64_t> kernel,
   return out;
 }
 
+torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
+                         c10::optional<torch::Tensor> bias, int64_t act) {
+  // x (M, K) bf16, w (N, K) bf16 (torch Linear layout) -> act(x@w^T+b)
+  TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
+  TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16);
+  const int m = (int)x.size(0), kk = (int)x.size(1), n = (int)w.size(0);
+  TORCH_CHECK(w.size(1) == kk && kk % 64 == 0 && n % 128 == 0,
+              "K%64==0 and N%128==0 required");
+  const void* bptr = nullptr;
+  torch::Tensor bc;
+  if (bias.has_value()) {
+    bc = bias->contiguous().to(torch::kBFloat16);
+    TORCH_CHECK(bc.numel() == n);
+    bptr = bc.data_ptr();
+  }
+  auto out = torch::empty({m, n}, x.options());
+  vfa_linear_act(x.data_ptr(), w.data_ptr(), bptr, out.data_ptr(), m, n, kk,
+                 (int)act, current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -394,5 +419,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("instance_norm2d", &instance_norm2d);
   m.def("maxpool3d_same", &maxpool3d_same);
   m.def("maxpool2d_same", &maxpool2d_same);
+  m.def("linear_act", &linear_act);
   m.attr("gfx_arch") = "gfx950";
 }
