@@ -1,7 +1,11 @@
 #!/usr/bin/env python3
 """Microbench: ops.linear_act (fused MFMA GEMM) vs torch F.linear(+act)
 on the CLIP / VGGish / RAFT shapes.  Run on a GPU box."""
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.join(os.path.dirname(__file__), '..'))
 
 import torch
 
