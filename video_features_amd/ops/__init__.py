@@ -347,7 +347,7 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
     n = weight.shape[0]
     if (_use_hip(x) and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and k % 64 == 0
-            and n % 128 == 0):
+            and n % 128 == 0 and not os.environ.get('VFA_NO_LTGEMM')):
         x2 = x.reshape(-1, k).contiguous()
         out = _ext.linear_act(x2, weight.contiguous(), bias, _ACT_IDS[act])
         return out.reshape(*x.shape[:-1], n)
