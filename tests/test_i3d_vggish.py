@@ -176,3 +176,32 @@ def test_i3d_flat_path_matches_5d_reference():
     assert out.shape == ref.shape == (2, 1024)
     assert torch.allclose(out, ref, atol=1e-4, rtol=1e-4), \
         (out - ref).abs().max().item()
+
+
+def test_i3d_weights_dir_loading(tmp_path):
+    """--weights_path as a directory of per-model checkpoints (incl.
+    DataParallel 'module.'-prefixed keys)."""
+    from video_features_amd.models.i3d import I3D
+    from video_features_amd.models.raft import RAFT
+    from video_features_amd.extractors.i3d import ExtractI3D
+    from video_features_amd.io.y4m import write_y4m
+    from tests.conftest import synthetic_frames
+
+    torch.manual_seed(7)
+    rgb_ref = I3D(modality='rgb')
+    raft_ref = RAFT()
+    torch.save(rgb_ref.state_dict(), tmp_path / 'i3d_rgb.pt')
+    torch.save({'module.' + k: v for k, v in raft_ref.state_dict().items()},
+               tmp_path / 'raft.pth')
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=12, h=64, w=64), fps=25.0)
+    cfg = Config(feature_type='i3d', video_paths=[vid], cpu=True,
+                 stack_size=10, step_size=10, flow_type='raft',
+                 weights_path=str(tmp_path))
+    ex = ExtractI3D(cfg, external_call=True)
+    models = ex.models_for(torch.device('cpu'))
+    # rgb stream got the saved weights (BN folded afterwards, so compare a
+    # conv weight that folding rescales only when BN stats are nontrivial)
+    assert torch.allclose(
+        models['flow_xtr'].update_block.flow_head.conv2.weight,
+        raft_ref.update_block.flow_head.conv2.weight)

@@ -41,19 +41,47 @@ class ExtractI3D(BaseExtractor):
         self.flow_type = self.cfg.flow_type
         self.output_feat_keys = list(self.streams)
 
+    def _maybe_load(self, model: torch.nn.Module, name: str) -> None:
+        """``--weights_path`` for i3d is a DIRECTORY holding
+        ``i3d_rgb.pt`` / ``i3d_flow.pt`` / ``raft.pth`` / ``pwc.pth``
+        (the reference ships these as separate checkpoint files,
+        reference extract_i3d.py:23-26); missing files stay random-init.
+        ``module.``-prefixed (DataParallel-saved) keys are accepted."""
+        wp = self.cfg.weights_path
+        if not wp:
+            return
+        p = Path(wp)
+        if p.is_dir():
+            p = p / f'{name}.pt'
+            if not p.exists():
+                p = p.with_suffix('.pth')
+        if not p.exists():
+            return
+        sd = torch.load(str(p), map_location='cpu', weights_only=True)
+        sd = {k.removeprefix('module.'): v for k, v in sd.items()}
+        model.load_state_dict(sd)
+
     def build_models(self, device: torch.device, dtype: torch.dtype):
         models = {}
         if 'rgb' in self.streams:
-            models['rgb'] = I3D(modality='rgb').to(device, dtype).eval()
+            rgb = I3D(modality='rgb')
+            self._maybe_load(rgb, 'i3d_rgb')
+            models['rgb'] = rgb.to(device, dtype).eval()
         if 'flow' in self.streams:
-            models['flow'] = I3D(modality='flow').to(device, dtype).eval()
+            flow = I3D(modality='flow')
+            self._maybe_load(flow, 'i3d_flow')
+            models['flow'] = flow.to(device, dtype).eval()
             if self.flow_type == 'raft':
-                raft = RAFT().to(device, dtype).eval()
+                raft = RAFT()
+                self._maybe_load(raft, 'raft')
+                raft = raft.to(device, dtype).eval()
                 if device.type == 'cuda':
                     raft = raft.use_channels_last()
                 models['flow_xtr'] = raft
             elif self.flow_type == 'pwc':
-                models['flow_xtr'] = PWCNet().to(device, dtype).eval()
+                pwc = PWCNet()
+                self._maybe_load(pwc, 'pwc')
+                models['flow_xtr'] = pwc.to(device, dtype).eval()
         return models
 
     # ------------------------------------------------------------ helpers
