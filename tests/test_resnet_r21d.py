@@ -64,3 +64,31 @@ def test_show_pred_labels():
     from video_features_amd.utils.labels import class_names
     assert len(class_names('kinetics')) == 400
     assert len(class_names('imagenet')) == 1000
+
+
+def test_r21d_flat_path_matches_5d_reference():
+    """forward_features (flattened-time, no conv3d) must match the plain
+    conv3d composition exactly."""
+    from video_features_amd.models.r21d import R2Plus1D18
+
+    torch.manual_seed(5)
+    model = R2Plus1D18().eval()
+    for m in model.modules():
+        if isinstance(m, torch.nn.BatchNorm3d):
+            m.running_mean.normal_(0, 0.3)
+            m.running_var.uniform_(0.5, 2.0)
+    x = torch.randn(2, 3, 16, 64, 64)
+
+    def ref(m, x):
+        x = m.stem(x)
+        for layer in (m.layer1, m.layer2, m.layer3, m.layer4):
+            for blk in layer:
+                x = blk.forward(x)
+        return m.avgpool(x).flatten(1)
+
+    with torch.no_grad():
+        out = model.forward_features(x)
+        expect = ref(model, x)
+    assert out.shape == expect.shape == (2, 512)
+    assert torch.allclose(out, expect, atol=2e-4, rtol=1e-4), \
+        (out - expect).abs().max().item()

@@ -44,80 +44,15 @@ def _triple(v) -> Tuple[int, int, int]:
 
 
 # ----------------------------------------------------- flattened-time path
-# On MI355X, MIOpen lowers NCDHW conv3d through Im3d2Col + GEMM (profiled in
-# profiles/) — an im2col materialization 27x the input for 3x3x3 kernels.
-# Instead the whole backbone below the stem runs on (B*T, C, H, W)
-# channels_last tensors: 1x1x1 convs become plain GEMM-shaped conv2d, a
-# 3x3x3 conv becomes ONE merged conv2d with 3*O outputs (the three temporal
-# taps) followed by a shifted add over t, and the TF-SAME 3D pools separate
-# exactly into a spatial 2D pool + a shifted temporal maximum (max over a
-# separable window commutes).
-
-def flatten_time(x: torch.Tensor) -> torch.Tensor:
-    """(B, C, T, H, W) → (B*T, C, H, W) channels_last, one copy."""
-    b, c, t, h, w = x.shape
-    y = x.permute(0, 2, 3, 4, 1).reshape(b * t, h, w, c)
-    return y.permute(0, 3, 1, 2)
-
-
-def unflatten_time(xf: torch.Tensor, b: int) -> torch.Tensor:
-    """(B*T, C, H, W) channels_last → (B, C, T, H, W) contiguous."""
-    bt, c, h, w = xf.shape
-    t = bt // b
-    return xf.permute(0, 2, 3, 1).reshape(b, t, h, w, c) \
-        .permute(0, 4, 1, 2, 3).contiguous()
-
-
-def _cl_empty(bt: int, c: int, h: int, w: int, like: torch.Tensor):
-    return torch.empty(bt, c, h, w, device=like.device, dtype=like.dtype,
-                       memory_format=torch.channels_last)
+# See models/_flat3d.py: the backbone below the stem runs on (B*T, C, H, W)
+# channels_last tensors (no conv3d, no Im3d2Col).
+from ._flat3d import (flatten_time, unflatten_time, cl_empty,
+                      temporal_merge, temporal_max)
 
 
 def temporal_merge3(y: torch.Tensor, b: int) -> torch.Tensor:
-    """y (B*T, 3O, H, W): the three temporal-tap conv outputs stacked along
-    channels → (B*T, O, H, W) with out[t] = y0[t-1] + y1[t] + y2[t+1]
-    (zero temporal TF-SAME padding; bias was applied on the middle tap)."""
-    bt, c3, h, w = y.shape
-    o = c3 // 3
-    t = bt // b
-    out = _cl_empty(bt, o, h, w, y)
-    out.copy_(y[:, o:2 * o])
-    y5 = y.view(b, t, c3, h, w)
-    o5 = out.view(b, t, o, h, w)
-    if t > 1:
-        o5[:, 1:] += y5[:, :-1, :o]
-        o5[:, :-1] += y5[:, 1:, 2 * o:]
-    return out
-
-
-def temporal_max(xf: torch.Tensor, b: int, kt: int, st: int) -> torch.Tensor:
-    """Shifted maximum along t with TF-SAME zero padding:
-    (B*T, C, H, W) → (B*T', C, H, W), T' = ceil(T/st)."""
-    bt, c, h, w = xf.shape
-    t = bt // b
-    p0, p1 = _same_pad_1d(t, kt, st)
-    to = (t + p0 + p1 - kt) // st + 1
-    out = _cl_empty(b * to, c, h, w, xf)
-    out.fill_(float('-inf'))
-    x5 = xf.view(b, t, c, h, w)
-    o5 = out.view(b, to, c, h, w)
-    for dt in range(kt):
-        j_lo = max(0, -(-(p0 - dt) // st))
-        j_hi = min(to - 1, (t - 1 + p0 - dt) // st)
-        if j_lo > j_hi:
-            continue
-        s_lo = j_lo * st - p0 + dt
-        src = x5[:, s_lo:s_lo + (j_hi - j_lo) * st + 1:st]
-        dst = o5[:, j_lo:j_hi + 1]
-        torch.maximum(dst, src, out=dst)
-    # zero padding participates in the max at the temporal edges
-    head = -(-p0 // st)                       # windows missing leading taps
-    if head > 0:
-        o5[:, :head].clamp_(min=0)
-    j_tail = -(-(t + p0 - kt + 1) // st)      # windows past the last frame
-    if j_tail < to:
-        o5[:, j_tail:].clamp_(min=0)
-    return out
+    """3-tap stride-1 merge (3x3x3 conv, TF-SAME temporal pad (1,1))."""
+    return temporal_merge(y, b, kt=3, st=1, p0=1, bias_tap=1)
 
 
 class Unit3D(nn.Module):
@@ -200,7 +135,9 @@ class MaxPool3dSame(nn.Module):
         y = ops.maxpool2d_same(xf, self.kernel[1:], self.stride[1:],
                                nhwc=nhwc)
         if self.kernel[0] > 1 or self.stride[0] > 1:
-            y = temporal_max(y, b, self.kernel[0], self.stride[0])
+            t = y.shape[0] // b
+            p0, p1 = _same_pad_1d(t, self.kernel[0], self.stride[0])
+            y = temporal_max(y, b, self.kernel[0], self.stride[0], p0, p1)
         return y
 
 

@@ -13,7 +13,17 @@ M chosen so the parameter count matches the full 3D conv:
 from __future__ import annotations
 
 import torch
+import torch.nn.functional as F
 from torch import nn
+
+from ._flat3d import (flatten_time, temporal_merge, temporal_select)
+
+
+def _flat_bn_relu(x, bn, relu: bool):
+    if isinstance(bn, nn.BatchNorm3d):
+        x = F.batch_norm(x, bn.running_mean, bn.running_var, bn.weight,
+                         bn.bias, bn.training, bn.momentum, bn.eps)
+    return F.relu(x, inplace=True) if relu else x
 
 
 def _midplanes(in_planes: int, out_planes: int, t: int = 3, k: int = 3) -> int:
@@ -35,6 +45,26 @@ class Conv2Plus1D(nn.Module):
 
     def forward(self, x):
         return self.temporal(self.relu(self.bn(self.spatial(x))))
+
+    def forward_flat(self, xf, b):
+        """(B*T, C, H, W) channels_last path (see models/_flat3d.py):
+        spatial (1,3,3) conv = conv2d; temporal (3,1,1) conv = merged
+        3-tap 1x1 conv2d + shifted (strided) temporal add."""
+        ss = self.spatial.stride[1]
+        y = F.conv2d(xf, self.spatial.weight[:, :, 0], self.spatial.bias,
+                     stride=ss, padding=1)
+        y = _flat_bn_relu(y, self.bn, True)
+        w = self.temporal.weight                   # (O, M, 3, 1, 1)
+        o = w.shape[0]
+        wcat = w.permute(2, 0, 1, 3, 4).reshape(3 * o, w.shape[1], 1, 1)
+        bias = self.temporal.bias
+        if bias is not None:
+            bcat = torch.zeros(3 * o, device=bias.device, dtype=bias.dtype)
+            bcat[o:2 * o] = bias
+        else:
+            bcat = None
+        y = F.conv2d(y, wcat, bcat)
+        return temporal_merge(y, b, kt=3, st=self.temporal.stride[0], p0=1)
 
 
 class R21DBlock(nn.Module):
@@ -61,6 +91,23 @@ class R21DBlock(nn.Module):
         out = self.bn2(self.conv2(out))
         return self.relu(out + identity)
 
+    def forward_flat(self, xf, b):
+        if self.downsample is None:
+            identity = xf
+        else:
+            conv, bn = self.downsample[0], self.downsample[1]
+            st = conv.stride[0]
+            # 1x1x1 stride-s conv: temporal subsample, then conv2d stride s
+            identity = temporal_select(xf, b, st)
+            identity = F.conv2d(identity, conv.weight[:, :, 0], conv.bias,
+                                stride=conv.stride[1])
+            identity = _flat_bn_relu(identity, bn, False)
+        out = self.conv1.forward_flat(xf, b)
+        out = _flat_bn_relu(out, self.bn1, True)
+        out = self.conv2.forward_flat(out, b)
+        out = _flat_bn_relu(out, self.bn2, False)
+        return F.relu(out + identity, inplace=True)
+
 
 class R2Plus1D18(nn.Module):
     def __init__(self, num_classes: int = 400):
@@ -80,10 +127,34 @@ class R2Plus1D18(nn.Module):
         self.fc = nn.Linear(512, num_classes)
 
     def forward_features(self, x: torch.Tensor) -> torch.Tensor:
-        """(B, 3, T, H, W) → (B, 512)."""
-        x = self.stem(x)
-        x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
-        return self.avgpool(x).flatten(1)
+        """(B, 3, T, H, W) → (B, 512) via the flattened-time path — the
+        R(2+1)D factorization is fully separable, so NO conv3d runs at
+        all (reference uses torchvision's NCDHW conv3d stack)."""
+        b = x.shape[0]
+        xf = flatten_time(x)
+        # stem: (1,7,7)/s(1,2,2) conv2d -> BN+ReLU -> 3-tap temporal 1x1
+        sc0, sbn0, sc1, sbn1 = (self.stem[0], self.stem[1], self.stem[3],
+                                self.stem[4])
+        xf = F.conv2d(xf, sc0.weight[:, :, 0], sc0.bias, stride=2, padding=3)
+        xf = _flat_bn_relu(xf, sbn0, True)
+        o = sc1.weight.shape[0]
+        wcat = sc1.weight.permute(2, 0, 1, 3, 4).reshape(3 * o,
+                                                         sc1.weight.shape[1],
+                                                         1, 1)
+        if sc1.bias is not None:
+            bcat = torch.zeros(3 * o, device=sc1.bias.device,
+                               dtype=sc1.bias.dtype)
+            bcat[o:2 * o] = sc1.bias
+        else:
+            bcat = None
+        xf = temporal_merge(F.conv2d(xf, wcat, bcat), b, kt=3, st=1, p0=1)
+        xf = _flat_bn_relu(xf, sbn1, True)
+        for layer in (self.layer1, self.layer2, self.layer3, self.layer4):
+            for blk in layer:
+                xf = blk.forward_flat(xf, b)
+        # global average over (T, H, W)
+        bt, c, h, w = xf.shape
+        return xf.view(b, bt // b, c, h, w).mean(dim=(1, 3, 4))
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.fc(self.forward_features(x))
