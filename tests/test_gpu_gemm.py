@@ -32,7 +32,8 @@ def _ref(x, w, b, act):
 
 @pytest.mark.parametrize('m,n,k', [(256, 128, 64), (512, 384, 192),
                                    (1000, 768, 3072), (9600, 3072, 768),
-                                   (129, 256, 128)])
+                                   (129, 256, 128), (333, 100, 72),
+                                   (2048, 2048, 520), (770, 530, 336)])
 @pytest.mark.parametrize('act', ['none', 'relu', 'quick_gelu', 'gelu'])
 def test_linear_act(dev, m, n, k, act):
     ops = _hip_loaded()

@@ -346,8 +346,8 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
     k = x.shape[-1]
     n = weight.shape[0]
     if (_use_hip(x) and x.dtype == torch.bfloat16
-            and weight.dtype == torch.bfloat16 and k % 64 == 0
-            and n % 128 == 0 and not os.environ.get('VFA_NO_LTGEMM')):
+            and weight.dtype == torch.bfloat16 and k % 8 == 0 and n >= 16
+            and not os.environ.get('VFA_NO_LTGEMM')):
         x2 = x.reshape(-1, k).contiguous()
         out = _ext.linear_act(x2, weight.contiguous(), bias, _ACT_IDS[act])
         return out.reshape(*x.shape[:-1], n)

@@ -383,8 +383,9 @@ torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
   TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
               w.scalar_type() == torch::kBFloat16);
   const int m = (int)x.size(0), kk = (int)x.size(1), n = (int)w.size(0);
-  TORCH_CHECK(w.size(1) == kk && kk % 64 == 0 && n % 128 == 0,
-              "K%64==0 and N%128==0 required");
+  // ragged M/N/K handled by the guarded staging path; K%8 keeps the
+  // 16-B row segments aligned
+  TORCH_CHECK(w.size(1) == kk && kk % 8 == 0, "K%8==0 required");
   const void* bptr = nullptr;
   torch::Tensor bc;
   if (bias.has_value()) {

@@ -5,19 +5,18 @@
 //   the epilogue (QuickGELU / tanh-GELU / ReLU / none) — the separate
 //   activation kernel's full HBM round trip disappears.
 //
-// Structure (cdna_hip_programming.md §5 ladder step 3 + st_16x32 swizzle):
-//   128x128 tile, BK = 64, 4 waves (2M x 2N), 64x64 C per wave as a 4x4
-//   grid of v_mfma_f32_16x16x32_bf16 accumulators.
-//   Both operands stage through LDS as [row][64] bf16 images (A rows = m,
-//   B rows = n; the B fragment of C[m][n] = dot_k A[m][k] W[n][k] reads
-//   the same row-major image as A).  Staging uses
-//   __builtin_amdgcn_global_load_lds width 16 (2 LDS buffers, next K-tile
-//   in flight during compute), with the st_16x32 XOR swizzle
-//   (byte ^= ((byte>>9)&1)<<5 inside each 1024 B subtile) applied to the
-//   *global source* address so the LDS image stays lane-linear for glds;
-//   ds_read_b128 fragment reads apply the same XOR.
-//   Partial tiles (M tail) take a bounds-checked vector-staging path with
-//   identical LDS image.
+// Structure (cdna_hip_programming.md §5): BIG tiles 256x256 (8 waves as
+// 2M x 4N, 128x64 C per wave, 128 KiB LDS) for the transformer shapes,
+// 128x128 (4 waves, 2x2) when N < 256 or M is small.  BK = 64.  Both
+// operands stage through LDS as [row][64] bf16 images (A rows = m, B rows
+// = n; the B fragment of C[m][n] = dot_k A[m][k] W[n][k] reads the same
+// row-major image as A).  Staging uses __builtin_amdgcn_global_load_lds
+// width 16 (2 LDS buffers, next K-tile in flight during compute), with the
+// st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5 inside each 1024 B
+// subtile) applied to the *global source* address so the LDS image stays
+// lane-linear for glds; ds_read_b128 fragment reads apply the same XOR.
+// Partial tiles (M/N/K tails) take a bounds-checked vector-staging path
+// with an identical LDS image.
 #include "vfa_common.h"
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
@@ -25,9 +24,7 @@ typedef float f32x4 __attribute__((ext_vector_type(4)));
 
 namespace {
 
-constexpr int BM = 128, BN = 128, BK = 64;
-constexpr int TILE_B = BM * BK * 2;            // bytes per operand tile
-constexpr int NSUB = TILE_B / 1024;            // 16 glds subtiles / operand
+constexpr int BK = 64;
 
 __device__ __forceinline__ int swz(int byte_off) {
   return byte_off ^ (((byte_off >> 9) & 1) << 5);
@@ -43,19 +40,21 @@ __device__ __forceinline__ float act_f(float x, int kind) {
   return x;
 }
 
-// stage a (rows x 64) bf16 tile into an LDS image with the st_16x32
-// swizzle via glds: each wave covers NSUB/4 1024-B subtiles (8 rows each).
+// stage a (ROWS x 64) bf16 tile into an LDS image with the st_16x32
+// swizzle via glds; each wave covers (ROWS*128/1024)/WAVES subtiles.
+template <int ROWS, int WAVES>
 __device__ __forceinline__ void stage_glds(const __bf16* __restrict__ g,
                                            long long row_stride_elems,
                                            char* lds_base, int wave,
                                            int lane) {
+  constexpr int NSUB = ROWS * 128 / 1024;
   const int off = lane * 16;                   // linear LDS offset in subtile
   const int off_log = off ^ (((off >> 9) & 1) << 5);
   const int r_in = off_log >> 7;               // row within subtile
   const int b_in = off_log & 127;              // byte within 128-B row
 #pragma unroll
-  for (int i = 0; i < NSUB / 4; ++i) {
-    const int sub = wave * (NSUB / 4) + i;
+  for (int i = 0; i < NSUB / WAVES; ++i) {
+    const int sub = wave * (NSUB / WAVES) + i;
     const __bf16* src = g + (long long)(sub * 8 + r_in) * row_stride_elems;
     // LDS destination is wave-uniform base + lane*16 (hardware-added);
     // the per-lane *global* address carries the swizzle
@@ -66,32 +65,51 @@ __device__ __forceinline__ void stage_glds(const __bf16* __restrict__ g,
   }
 }
 
-// bounds-checked fallback staging (M tail): same LDS image, zero fill.
+// bounds-checked fallback staging (tails): same LDS image, zero fill.
+template <int ROWS, int THREADS>
 __device__ __forceinline__ void stage_guard(const __bf16* __restrict__ g,
                                             long long row_stride_elems,
-                                            int valid_rows, char* lds_base,
-                                            int tid) {
-  for (int t = tid; t < BM * 8; t += 256) {    // 8 x 16-B segments per row
+                                            int valid_rows, int valid_k,
+                                            char* lds_base, int tid) {
+  for (int t = tid; t < ROWS * 8; t += THREADS) {   // 8 16-B segs per row
     const int row = t >> 3, seg = t & 7;
     uint4 v = {0u, 0u, 0u, 0u};
-    if (row < valid_rows)
-      v = *reinterpret_cast<const uint4*>(g + row * row_stride_elems +
-                                          seg * 8);
+    if (row < valid_rows && seg * 8 < valid_k) {
+      if ((seg + 1) * 8 <= valid_k) {
+        v = *reinterpret_cast<const uint4*>(g + row * row_stride_elems +
+                                            seg * 8);
+      } else {
+        __bf16* e = reinterpret_cast<__bf16*>(&v);
+        for (int j = 0; seg * 8 + j < valid_k; ++j)
+          e[j] = g[row * row_stride_elems + seg * 8 + j];
+      }
+    }
     *reinterpret_cast<uint4*>(lds_base + swz(row * 128 + seg * 16)) = v;
   }
 }
 
-template <int ACT, bool FULL>
-__global__ __launch_bounds__(256)
+// BIG: 256x256 tile, 8 waves (2Mx4N), 128x64 C per wave (8x4 MFMA tiles).
+// else: 128x128 tile, 4 waves (2x2), 64x64 per wave (4x4 tiles).
+template <int ACT, bool FULL, bool BIG>
+__global__ __launch_bounds__(BIG ? 512 : 256)
 void linear_act_kernel(const __bf16* __restrict__ a,
                        const __bf16* __restrict__ w,
                        const __bf16* __restrict__ bias,
                        __bf16* __restrict__ c, int m, int n, int k,
                        int tiles_m, int tiles_n) {
+  constexpr int BM = BIG ? 256 : 128, BN = BIG ? 256 : 128;
+  constexpr int WAVES = BIG ? 8 : 4;
+  constexpr int WN = BIG ? 4 : 2;              // waves along N
+  constexpr int MI = BIG ? 8 : 4;              // 16-row MFMA tiles per wave
+  constexpr int NJ = 4;                        // 16-col MFMA tiles per wave
+  constexpr int TILE_A = BM * BK * 2, TILE_BB = BN * BK * 2;
+  constexpr int THREADS = BIG ? 512 : 256;
+
   extern __shared__ __attribute__((aligned(1024))) char smem[];
-  // layout: [buf][A | B], each TILE_B bytes
-  auto sA = [&](int buf) { return smem + buf * 2 * TILE_B; };
-  auto sB = [&](int buf) { return smem + buf * 2 * TILE_B + TILE_B; };
+  auto sA = [&](int buf) { return smem + buf * (TILE_A + TILE_BB); };
+  auto sB = [&](int buf) {
+    return smem + buf * (TILE_A + TILE_BB) + TILE_A;
+  };
 
   // XCD-aware bijective remap (consecutive remapped ids share an XCD)
   const int nwg = tiles_m * tiles_n;
@@ -106,62 +124,59 @@ void linear_act_kernel(const __bf16* __restrict__ a,
 
   const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   const int lo = lane & 15, hi4 = lane >> 4;
-  const int wm = wave >> 1, wn = wave & 1;     // 2x2 wave grid
+  const int wm = wave / WN, wn = wave % WN;
 
-  const int nk = k / BK;
-  const int valid_m = m - m0;
+  const int nk = (k + BK - 1) / BK;
+  const int valid_m = m - m0, valid_n = n - n0;
 
-  if (FULL) {
-    stage_glds(a + (long long)m0 * k, k, sA(0), wave, lane);
-    stage_glds(w + (long long)n0 * k, k, sB(0), wave, lane);
-  } else {
-    stage_guard(a + (long long)m0 * k, k, valid_m, sA(0), threadIdx.x);
-    stage_guard(w + (long long)n0 * k, k, n - n0, sB(0), threadIdx.x);
-  }
+  auto stage = [&](int buf, int kt) {
+    const long long ko = (long long)kt * BK;
+    if (FULL) {
+      stage_glds<BM, WAVES>(a + (long long)m0 * k + ko, k, sA(buf), wave,
+                            lane);
+      stage_glds<BN, WAVES>(w + (long long)n0 * k + ko, k, sB(buf), wave,
+                            lane);
+    } else {
+      const int vk = (int)min((long long)BK, (long long)k - ko);
+      stage_guard<BM, THREADS>(a + (long long)m0 * k + ko, k, valid_m, vk,
+                               sA(buf), threadIdx.x);
+      stage_guard<BN, THREADS>(w + (long long)n0 * k + ko, k, valid_n, vk,
+                               sB(buf), threadIdx.x);
+    }
+  };
 
-  f32x4 acc[4][4];
+  stage(0, 0);
+
+  f32x4 acc[MI][NJ];
 #pragma unroll
-  for (int i = 0; i < 4; ++i)
+  for (int i = 0; i < MI; ++i)
 #pragma unroll
-    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+    for (int j = 0; j < NJ; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
 
   for (int kt = 0; kt < nk; ++kt) {
     __syncthreads();                           // tile kt resident
     const int cur = kt & 1;
-    if (kt + 1 < nk) {
-      const int nxt = 1 - cur;
-      if (FULL) {
-        stage_glds(a + (long long)m0 * k + (kt + 1) * BK, k, sA(nxt), wave,
-                   lane);
-        stage_glds(w + (long long)n0 * k + (kt + 1) * BK, k, sB(nxt), wave,
-                   lane);
-      } else {
-        stage_guard(a + (long long)m0 * k + (kt + 1) * BK, k, valid_m,
-                    sA(nxt), threadIdx.x);
-        stage_guard(w + (long long)n0 * k + (kt + 1) * BK, k, n - n0,
-                    sB(nxt), threadIdx.x);
-      }
-    }
+    if (kt + 1 < nk) stage(1 - cur, kt + 1);
     // compute on tile kt
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {           // two 32-deep MFMA steps
-      bf16x8 afr[4], bfr[4];
+      bf16x8 afr[MI], bfr[NJ];
 #pragma unroll
-      for (int i = 0; i < 4; ++i) {
-        const int arow = wm * 64 + i * 16 + lo;
-        afr[i] = *reinterpret_cast<const bf16x8*>(
-            sA(cur) + swz(arow * 128 + kk * 64 + hi4 * 16));
-      }
-#pragma unroll
-      for (int j = 0; j < 4; ++j) {
-        const int brow = wn * 64 + j * 16 + lo;
+      for (int j = 0; j < NJ; ++j) {
+        const int brow = wn * (NJ * 16) + j * 16 + lo;
         bfr[j] = *reinterpret_cast<const bf16x8*>(
             sB(cur) + swz(brow * 128 + kk * 64 + hi4 * 16));
       }
 #pragma unroll
-      for (int i = 0; i < 4; ++i)
+      for (int i = 0; i < MI; ++i) {
+        const int arow = wm * (MI * 16) + i * 16 + lo;
+        afr[i] = *reinterpret_cast<const bf16x8*>(
+            sA(cur) + swz(arow * 128 + kk * 64 + hi4 * 16));
+      }
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
+      for (int i = 0; i < MI; ++i)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afr[i], bfr[j], acc[i][j], 0, 0, 0);
     }
@@ -171,14 +186,15 @@ void linear_act_kernel(const __bf16* __restrict__ a,
 
   // epilogue: bias + activation, bf16 store
 #pragma unroll
-  for (int j = 0; j < 4; ++j) {
-    const int col = n0 + wn * 64 + j * 16 + lo;
+  for (int j = 0; j < NJ; ++j) {
+    const int col = n0 + wn * (NJ * 16) + j * 16 + lo;
+    if (!FULL && col >= n) continue;
     const float bv = bias ? (float)bias[col] : 0.f;
 #pragma unroll
-    for (int i = 0; i < 4; ++i) {
+    for (int i = 0; i < MI; ++i) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
-        const int row = m0 + wm * 64 + i * 16 + hi4 * 4 + r;
+        const int row = m0 + wm * (MI * 16) + i * 16 + hi4 * 4 + r;
         if (!FULL && row >= m) continue;
         c[(long long)row * n + col] =
             (__bf16)act_f(acc[i][j][r] + bv, ACT);
@@ -187,30 +203,41 @@ void linear_act_kernel(const __bf16* __restrict__ a,
   }
 }
 
+template <int ACT, bool BIG>
+void launch_tile(const void* a, const void* w, const void* bias, void* c,
+                 int m, int n, int k, hipStream_t stream) {
+  constexpr int BM = BIG ? 256 : 128, BN = BIG ? 256 : 128;
+  const int tiles_m = (m + BM - 1) / BM, tiles_n = (n + BN - 1) / BN;
+  const dim3 grid(tiles_m * tiles_n);
+  const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
+  const bool full = (m % BM == 0) && (n % BN == 0) && (k % BK == 0);
+  if (full)
+    hipLaunchKernelGGL((linear_act_kernel<ACT, true, BIG>), grid,
+                       dim3(BIG ? 512 : 256), lds, stream, (const __bf16*)a,
+                       (const __bf16*)w, (const __bf16*)bias, (__bf16*)c, m,
+                       n, k, tiles_m, tiles_n);
+  else
+    hipLaunchKernelGGL((linear_act_kernel<ACT, false, BIG>), grid,
+                       dim3(BIG ? 512 : 256), lds, stream, (const __bf16*)a,
+                       (const __bf16*)w, (const __bf16*)bias, (__bf16*)c, m,
+                       n, k, tiles_m, tiles_n);
+}
+
 template <int ACT>
 void launch_linear(const void* a, const void* w, const void* bias, void* c,
                    int m, int n, int k, hipStream_t stream) {
-  const int tiles_m = (m + BM - 1) / BM, tiles_n = n / BN;
-  const dim3 grid(tiles_m * tiles_n);
-  const size_t lds = 4 * TILE_B;               // 2 buffers x (A + B)
-  if (m % BM == 0)
-    hipLaunchKernelGGL((linear_act_kernel<ACT, true>), grid, dim3(256), lds,
-                       stream, (const __bf16*)a, (const __bf16*)w,
-                       (const __bf16*)bias, (__bf16*)c, m, n, k, tiles_m,
-                       tiles_n);
+  // BIG tiles when the problem can fill the chip with them
+  if (n >= 256 && (long long)((m + 255) / 256) * ((n + 255) / 256) >= 150)
+    launch_tile<ACT, true>(a, w, bias, c, m, n, k, stream);
   else
-    hipLaunchKernelGGL((linear_act_kernel<ACT, false>), grid, dim3(256), lds,
-                       stream, (const __bf16*)a, (const __bf16*)w,
-                       (const __bf16*)bias, (__bf16*)c, m, n, k, tiles_m,
-                       tiles_n);
+    launch_tile<ACT, false>(a, w, bias, c, m, n, k, stream);
 }
 
 }  // namespace
 
 extern "C" {
 
-// act: 0 none, 1 relu, 2 quick_gelu, 3 gelu_tanh.  Requires K % 64 == 0,
-// N % 128 == 0 (checked at the binding).
+// act: 0 none, 1 relu, 2 quick_gelu, 3 gelu_tanh
 void vfa_linear_act(const void* a, const void* w, const void* bias, void* c,
                     int m, int n, int k, int act, hipStream_t stream) {
   switch (act) {
