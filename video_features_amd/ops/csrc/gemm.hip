@@ -128,10 +128,13 @@ void linear_act_kernel(const __bf16* __restrict__ a,
 
   const int nk = (k + BK - 1) / BK;
   const int valid_m = m - m0, valid_n = n - n0;
+  // interior tiles take the glds fast path even when the problem has
+  // edge tiles; FULL means k % BK == 0 (no ragged K-step anywhere)
+  const bool tile_full = valid_m >= BM && valid_n >= BN;
 
   auto stage = [&](int buf, int kt) {
     const long long ko = (long long)kt * BK;
-    if (FULL) {
+    if (tile_full && (FULL || kt + 1 < nk)) {
       stage_glds<BM, WAVES>(a + (long long)m0 * k + ko, k, sA(buf), wave,
                             lane);
       stage_glds<BN, WAVES>(w + (long long)n0 * k + ko, k, sB(buf), wave,
@@ -184,18 +187,18 @@ void linear_act_kernel(const __bf16* __restrict__ a,
     // drains the in-flight glds and orders reads before buffer reuse
   }
 
-  // epilogue: bias + activation, bf16 store
+  // epilogue: bias + activation, bf16 store (bounds only on edge tiles)
 #pragma unroll
   for (int j = 0; j < NJ; ++j) {
     const int col = n0 + wn * (NJ * 16) + j * 16 + lo;
-    if (!FULL && col >= n) continue;
+    if (!tile_full && col >= n) continue;
     const float bv = bias ? (float)bias[col] : 0.f;
 #pragma unroll
     for (int i = 0; i < MI; ++i) {
 #pragma unroll
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm * (MI * 16) + i * 16 + hi4 * 4 + r;
-        if (!FULL && row >= m) continue;
+        if (!tile_full && row >= m) continue;
         c[(long long)row * n + col] =
             (__bf16)act_f(acc[i][j][r] + bv, ACT);
       }
@@ -210,7 +213,7 @@ void launch_tile(const void* a, const void* w, const void* bias, void* c,
   const int tiles_m = (m + BM - 1) / BM, tiles_n = (n + BN - 1) / BN;
   const dim3 grid(tiles_m * tiles_n);
   const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
-  const bool full = (m % BM == 0) && (n % BN == 0) && (k % BK == 0);
+  const bool full = (k % BK == 0);   // per-tile M/N edges handled inside
   if (full)
     hipLaunchKernelGGL((linear_act_kernel<ACT, true, BIG>), grid,
                        dim3(BIG ? 512 : 256), lds, stream, (const __bf16*)a,
