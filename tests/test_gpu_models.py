@@ -1,0 +1,93 @@
+"""GPU forwards of the non-CLIP model families vs their CPU fp32
+references (same weights, same inputs)."""
+import pytest
+import torch
+
+pytestmark = pytest.mark.gpu
+
+
+@pytest.fixture(scope='module')
+def dev():
+    assert torch.cuda.is_available()
+    return torch.device('cuda:0')
+
+
+def _cos(a, b):
+    return torch.nn.functional.cosine_similarity(
+        a.flatten().float(), b.flatten().float(), dim=0).item()
+
+
+def test_r21d_gpu_vs_cpu(dev):
+    from video_features_amd.models.r21d import R2Plus1D18
+    torch.manual_seed(0)
+    m = R2Plus1D18().eval()
+    x = torch.randn(2, 3, 16, 112, 112)
+    with torch.no_grad():
+        ref = m.forward_features(x)
+        out = m.to(dev).forward_features(x.to(dev)).cpu()
+    assert out.shape == (2, 512)
+    assert _cos(out, ref) > 0.999
+
+
+def test_r21d_gpu_bf16(dev):
+    from video_features_amd.models.r21d import R2Plus1D18
+    torch.manual_seed(0)
+    m = R2Plus1D18().eval()
+    x = torch.randn(2, 3, 16, 112, 112)
+    with torch.no_grad():
+        ref = m.forward_features(x)
+        out = m.to(dev, torch.bfloat16) \
+            .forward_features(x.to(dev, torch.bfloat16)).float().cpu()
+    assert _cos(out, ref) > 0.98
+
+
+def test_i3d_gpu_vs_cpu(dev):
+    from video_features_amd.models.i3d import I3D
+    torch.manual_seed(0)
+    m = I3D(modality='rgb').eval()
+    x = torch.randn(1, 3, 16, 128, 128)
+    with torch.no_grad():
+        ref = m.forward_features(x)
+        out = m.to(dev).forward_features(x.to(dev)).cpu()
+    assert out.shape == (1, 1024)
+    assert _cos(out, ref) > 0.999
+
+
+def test_resnet50_gpu_vs_cpu(dev):
+    from video_features_amd.models.resnet import build_resnet
+    torch.manual_seed(0)
+    m = build_resnet('resnet50').eval()
+    x = torch.randn(4, 3, 224, 224)
+    with torch.no_grad():
+        ref = m.forward_features(x)
+        mm = m.to(dev).to(memory_format=torch.channels_last)
+        out = mm.forward_features(
+            x.to(dev).contiguous(memory_format=torch.channels_last)).cpu()
+    assert out.shape == (4, 2048)
+    assert _cos(out, ref) > 0.999
+
+
+def test_vggish_gpu_vs_cpu(dev):
+    from video_features_amd.models.vggish import VGGish, waveform_to_examples
+    torch.manual_seed(0)
+    m = VGGish().eval()
+    wav = torch.sin(torch.arange(16000 * 2) / 16000 * 2 * 3.14159 * 440)
+    ex = waveform_to_examples(wav, 16000)
+    with torch.no_grad():
+        ref = m(ex)
+        out = m.to(dev)(ex.to(dev)).cpu()
+    assert out.shape[1] == 128
+    assert _cos(out, ref) > 0.999
+
+
+def test_pwc_gpu_vs_cpu(dev):
+    from video_features_amd.models.pwc import PWCNet
+    torch.manual_seed(0)
+    m = PWCNet().eval()
+    x1 = torch.rand(2, 3, 128, 192) * 255
+    x2 = torch.rand(2, 3, 128, 192) * 255
+    with torch.no_grad():
+        ref = m(x1, x2)
+        out = m.to(dev)(x1.to(dev), x2.to(dev)).cpu()
+    assert out.shape == ref.shape == (2, 2, 128, 192)
+    assert _cos(out, ref) > 0.99
