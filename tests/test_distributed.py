@@ -68,3 +68,36 @@ def test_two_process_gather_features(tmp_path):
     for a, b in zip(feats, solo):
         np.testing.assert_allclose(a['CLIP-ViT-B/32'], b['CLIP-ViT-B/32'],
                                    rtol=1e-4, atol=1e-5)
+
+
+@pytest.mark.timeout(600)
+def test_temporal_parallel_matches_single(tmp_path):
+    """--temporal_parallel: one video's sliding windows sharded over 2
+    ranks must reproduce the single-process features exactly (windows are
+    independent; merge interleaves rank rows back in order)."""
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    from tests.conftest import synthetic_frames
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'long.y4m')
+    write_y4m(vid, synthetic_frames(t=45, h=64, w=64), fps=25.0)
+    base = Config(feature_type='i3d', video_paths=[vid], cpu=True,
+                  stack_size=10, step_size=10, flow_type='pwc',
+                  gather_features=True, tmp_path=str(tmp_path / 'tmp'))
+    solo = run_extraction(base)
+    tp = run_extraction(base.replace(device_ids=[0, 1],
+                                     temporal_parallel=True))
+    assert len(tp) == len(solo) == 1
+    assert tp[0]['rgb'].shape == solo[0]['rgb'].shape == (4, 1024)
+    np.testing.assert_allclose(tp[0]['timestamps_ms'],
+                               solo[0]['timestamps_ms'])
+    np.testing.assert_allclose(tp[0]['rgb'], solo[0]['rgb'],
+                               rtol=1e-4, atol=1e-4)
+    np.testing.assert_allclose(tp[0]['flow'], solo[0]['flow'],
+                               rtol=1e-4, atol=1e-4)
+
+
+def test_temporal_parallel_rejected_for_framewise():
+    from video_features_amd.config import sanity_check
+    with pytest.raises(ValueError):
+        sanity_check(Config(feature_type='resnet50',
+                            temporal_parallel=True))

@@ -47,6 +47,9 @@ def build_parser() -> argparse.ArgumentParser:
     # ---- additive (not in the reference)
     p.add_argument('--dtype', choices=['auto', 'fp32', 'bf16'], default='auto')
     p.add_argument('--gather_features', action='store_true', default=False)
+    p.add_argument('--temporal_parallel', action='store_true', default=False,
+                   help='shard each video\'s sliding windows across ranks '
+                        '(i3d / r21d_rgb); exact — windows are independent')
     p.add_argument('--resume', action='store_true', default=False,
                    help='skip videos whose outputs already exist')
     p.add_argument('--profile', action='store_true', default=False)

@@ -82,6 +82,13 @@ class Config:
     seed: int = 0                  # random-init weight seed (no-network setting)
     weights_path: Optional[str] = None  # optional state_dict file for the model
 
+    # ---- temporal (context) parallelism: shard ONE video's sliding
+    # windows across ranks — exact, windows are independent (the flow
+    # "halo" frame is decoded locally); the reference has no equivalent
+    temporal_parallel: bool = False
+    tp_rank: int = 0      # set by the runtime, not the CLI
+    tp_world: int = 1
+
     def __post_init__(self) -> None:
         if self.feature_type not in FEATURE_TYPES:
             raise ValueError(
@@ -127,6 +134,9 @@ def sanity_check(cfg: Config) -> None:
                              f'(both are {cfg.output_path!r})')
     if cfg.show_pred and cfg.device_ids and len(cfg.device_ids) > 1:
         raise ValueError('--show_pred supports a single device only')
+    if cfg.temporal_parallel and cfg.feature_type not in ('i3d', 'r21d_rgb'):
+        raise ValueError('--temporal_parallel applies to the stack-windowed '
+                         "extractors ('i3d', 'r21d_rgb') only")
     if cfg.feature_type == 'r21d_rgb' and cfg.extraction_fps is not None:
         raise ValueError('r21d_rgb does not support custom extraction_fps '
                          '(the pretrained R(2+1)D assumes native fps)')

@@ -142,6 +142,9 @@ class ExtractI3D(BaseExtractor):
         ssz, step = self.stack_size, self.step_size
         starts = [s for s in range(0, max(n - ssz, 1), step)
                   if s + ssz + 1 <= n]
+        # temporal parallelism: this rank owns every tp_world-th window
+        # (merged back in rank order by runtime.dist.merge_temporal_shards)
+        starts = starts[self.cfg.tp_rank::self.cfg.tp_world]
         bs = max(1, self.cfg.batch_size or 1)
         for i in range(0, len(starts), bs):
             grp = starts[i:i + bs]

@@ -44,6 +44,8 @@ class ExtractR21D(BaseExtractor):
         slices = form_slices(n, self.stack_size, self.step_size)
         if not slices:
             slices = [(0, n)]   # shorter than one stack: use what exists
+        # temporal parallelism: this rank owns every tp_world-th window
+        slices = slices[self.cfg.tp_rank::self.cfg.tp_world]
         dtype = self.compute_dtype(device)
         feats, ts = [], []
         for (start, end) in slices:

@@ -80,6 +80,26 @@ def shard_indices(n: int, rank: int, world: int) -> torch.LongTensor:
     return torch.arange(n, dtype=torch.long)[rank::world]
 
 
+def merge_temporal_shards(parts: List[Dict], meta_keys=('fps',)) -> Dict:
+    """Reassemble per-rank window shards of ONE video: rank r produced the
+    windows ``starts[r::world]`` in order, so merged row i comes from rank
+    ``i % world``, position ``i // world``."""
+    import numpy as np
+    world = len(parts)
+    out: Dict = {}
+    for key, v0 in parts[0].items():
+        arrs = [p[key] for p in parts]
+        if key in meta_keys or np.ndim(v0) == 0:
+            out[key] = v0
+            continue
+        total = sum(a.shape[0] for a in arrs)
+        merged = np.empty((total,) + tuple(np.shape(v0)[1:]), dtype=np.asarray(v0).dtype)
+        for r, a in enumerate(arrs):
+            merged[r::world] = a
+        out[key] = merged
+    return out
+
+
 def _worker(rank: int, world: int, devices: List[str], cfg: Config,
             port: int, return_dict) -> None:
     device = torch.device(devices[rank])
@@ -97,11 +117,43 @@ def _worker(rank: int, world: int, devices: List[str], cfg: Config,
                         if backend == 'nccl' else dist.group.WORLD)
     try:
         extractor_cls = get_extractor_class(cfg.feature_type)
-        extractor = extractor_cls(cfg, external_call=cfg.gather_features)
+        tp = cfg.temporal_parallel and world > 1
+        if tp:
+            # temporal parallelism: every rank runs EVERY video, owning its
+            # stride of the sliding windows; rank 0 merges and sinks
+            cfg = cfg.replace(tp_rank=rank, tp_world=world)
+        extractor = extractor_cls(cfg, external_call=cfg.gather_features or tp)
         models = extractor.models_for(device)
         broadcast_models(models, src=0)
-        idxs = shard_indices(len(extractor.path_list), rank, world).to(device)
+        if tp:
+            idxs = torch.arange(len(extractor.path_list),
+                                dtype=torch.long).to(device)
+        else:
+            idxs = shard_indices(len(extractor.path_list), rank,
+                                 world).to(device)
         feats_list = extractor(idxs)
+        if tp:
+            group = gather_group or (dist.new_group(backend='gloo')
+                                     if backend == 'nccl' else dist.group.WORLD)
+            gathered = [None] * world if rank == 0 else None
+            dist.gather_object(feats_list, gathered, dst=0, group=group)
+            if rank == 0:
+                from .sinks import action_on_extraction
+                merged_all = [merge_temporal_shards(
+                    [g[i] for g in gathered],
+                    meta_keys=('fps',))
+                    for i in range(len(feats_list))]
+                for i, feats in enumerate(merged_all):
+                    if cfg.gather_features:
+                        continue
+                    action_on_extraction(
+                        feats, extractor._stem_path(extractor.path_list[i]),
+                        extractor.output_path, cfg.on_extraction,
+                        cfg.output_direct, cfg.feature_type)
+                if cfg.gather_features and return_dict is not None:
+                    return_dict['features'] = merged_all
+            dist.barrier()
+            return
         if cfg.gather_features:
             shard = [(int(i), f) for i, f in zip(idxs.tolist(), feats_list)]
             gathered: Optional[List] = [None] * world if rank == 0 else None
