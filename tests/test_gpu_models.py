@@ -72,7 +72,7 @@ def test_vggish_gpu_vs_cpu(dev):
     torch.manual_seed(0)
     m = VGGish().eval()
     wav = torch.sin(torch.arange(16000 * 2) / 16000 * 2 * 3.14159 * 440)
-    ex = waveform_to_examples(wav, 16000)
+    ex = waveform_to_examples(wav)
     with torch.no_grad():
         ref = m(ex)
         out = m.to(dev)(ex.to(dev)).cpu()

@@ -325,6 +325,17 @@ def maxpool2d_same(x: torch.Tensor, kernel, stride,
     h, w = x.shape[-2:]
     ph, pw = _pad1(h, kernel[0], stride[0]), _pad1(w, kernel[1], stride[1])
     if _use_hip(x):
+        # resolve the layout defensively: a tensor can be CL-contiguous,
+        # NCHW-contiguous, both (trivial dims), or neither (strided view)
+        if nhwc and not x.is_contiguous(
+                memory_format=torch.channels_last):
+            x = x.contiguous(memory_format=torch.channels_last)
+        elif not nhwc and not x.is_contiguous():
+            if x.is_contiguous(memory_format=torch.channels_last):
+                nhwc = True
+            else:
+                x = x.contiguous(memory_format=torch.channels_last)
+                nhwc = True
         out_sz = [(h + ph[0] + ph[1] - kernel[0]) // stride[0] + 1,
                   (w + pw[0] + pw[1] - kernel[1]) // stride[1] + 1]
         return _ext.maxpool2d_same(x, list(kernel), list(stride),
