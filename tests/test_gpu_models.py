@@ -91,3 +91,25 @@ def test_pwc_gpu_vs_cpu(dev):
         out = m.to(dev)(x1.to(dev), x2.to(dev)).cpu()
     assert out.shape == ref.shape == (2, 2, 128, 192)
     assert _cos(out, ref) > 0.99
+
+
+def test_resnet50_fused_bottleneck_path(dev):
+    """BN-folded bf16 CL path (1x1 convs as fused MFMA GEMMs with
+    residual+ReLU epilogue) vs the folded CPU fp32 reference."""
+    from video_features_amd.models.resnet import build_resnet
+    from video_features_amd.utils.fold_bn import fold_batchnorms
+    torch.manual_seed(0)
+    m = build_resnet('resnet50').eval()
+    for mod in m.modules():
+        if isinstance(mod, torch.nn.BatchNorm2d):
+            mod.running_mean.normal_(0, 0.2)
+            mod.running_var.uniform_(0.5, 2.0)
+    fold_batchnorms(m)
+    x = torch.randn(4, 3, 128, 128)
+    with torch.no_grad():
+        ref = m.forward_features(x)
+        mm = m.to(dev, torch.bfloat16).to(memory_format=torch.channels_last)
+        out = mm.forward_features(
+            x.to(dev, torch.bfloat16)
+            .contiguous(memory_format=torch.channels_last)).float().cpu()
+    assert _cos(out, ref) > 0.99, _cos(out, ref)

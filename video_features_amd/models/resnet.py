@@ -14,7 +14,10 @@ from __future__ import annotations
 from typing import List, Type
 
 import torch
+import torch.nn.functional as F
 from torch import nn
+
+from .. import ops
 
 
 class BasicBlock(nn.Module):
@@ -59,7 +62,35 @@ class Bottleneck(nn.Module):
                 nn.Conv2d(in_planes, out_planes, 1, stride, bias=False),
                 nn.BatchNorm2d(out_planes))
 
+    def _fused_ready(self, x) -> bool:
+        # inference path: BNs folded into the convs, bf16 channels_last on
+        # GPU with the HIP extension present
+        return (isinstance(self.bn1, nn.Identity) and x.is_cuda
+                and x.dtype == torch.bfloat16
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and ops.hip_available()
+                and not __import__('os').environ.get('VFA_NO_LTGEMM'))
+
     def forward(self, x):
+        if self._fused_ready(x):
+            # 1x1 convs as fused MFMA GEMMs on the CL view: conv1+ReLU in
+            # one epilogue; conv3 + residual-add + ReLU in one epilogue
+            # (removes MIOpen's SubTensor zero-fill + the eager add/relu
+            # round trips — see profiles/)
+            idt = x if self.downsample is None else self.downsample(x)
+            if self.conv1.out_channels >= 128:
+                y = ops.conv1x1_act(x, self.conv1.weight, self.conv1.bias,
+                                    'relu')
+            else:
+                y = F.relu(F.conv2d(x, self.conv1.weight, self.conv1.bias),
+                           inplace=True)
+            y = F.relu(F.conv2d(y, self.conv2.weight, self.conv2.bias,
+                                stride=self.conv2.stride, padding=1),
+                       inplace=True)
+            if not idt.is_contiguous(memory_format=torch.channels_last):
+                idt = idt.contiguous(memory_format=torch.channels_last)
+            return ops.conv1x1_act(y, self.conv3.weight, self.conv3.bias,
+                                   'relu', res=idt)
         identity = x if self.downsample is None else self.downsample(x)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.relu(self.bn2(self.conv2(out)))

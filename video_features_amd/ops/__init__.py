@@ -349,20 +349,25 @@ _ACT_IDS = {'none': 0, 'relu': 1, 'quick_gelu': 2, 'gelu': 3}
 
 def linear_act(x: torch.Tensor, weight: torch.Tensor,
                bias: Optional[torch.Tensor] = None,
-               act: str = 'none') -> torch.Tensor:
-    """``act(x @ weight^T + bias)`` — one fused MFMA GEMM on GPU (bf16,
-    128x128 tiles, glds-staged LDS with st_16x32 swizzle, bias+activation
-    in the epilogue).  Falls back to F.linear + activation on CPU or for
-    shapes outside K%64==0 / N%128==0."""
+               act: str = 'none',
+               res: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """``act(x @ weight^T + bias [+ res])`` — one fused MFMA GEMM on GPU
+    (bf16, 128/256-wide tiles, glds-staged LDS with st_16x32 swizzle; bias,
+    residual add, and activation in the epilogue).  Falls back to F.linear
+    (+ add + activation) on CPU or unsupported shapes."""
     k = x.shape[-1]
     n = weight.shape[0]
     if (_use_hip(x) and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and k % 8 == 0 and n >= 16
             and not os.environ.get('VFA_NO_LTGEMM')):
         x2 = x.reshape(-1, k).contiguous()
-        out = _ext.linear_act(x2, weight.contiguous(), bias, _ACT_IDS[act])
+        r2 = res.reshape(-1, n).contiguous() if res is not None else None
+        out = _ext.linear_act(x2, weight.contiguous(), bias, r2,
+                              _ACT_IDS[act])
         return out.reshape(*x.shape[:-1], n)
     y = torch.nn.functional.linear(x, weight, bias)
+    if res is not None:
+        y = y + res.reshape(y.shape)
     if act == 'relu':
         return torch.nn.functional.relu(y)
     if act == 'quick_gelu':
@@ -370,6 +375,20 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
     if act == 'gelu':
         return torch.nn.functional.gelu(y, approximate='tanh')
     return y
+
+
+def conv1x1_act(x: torch.Tensor, weight: torch.Tensor,
+                bias: Optional[torch.Tensor] = None, act: str = 'none',
+                res: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """1x1 conv on a channels_last (B, C, H, W) tensor as ONE fused MFMA
+    GEMM (M = B*H*W) with optional residual + activation epilogue — zero
+    layout copies: the CL tensor IS the (M, C) matrix."""
+    b, cin, h, w = x.shape
+    x2 = x.permute(0, 2, 3, 1).reshape(-1, cin)
+    r2 = res.permute(0, 2, 3, 1).reshape(-1, weight.shape[0]) \
+        if res is not None else None
+    y = linear_act(x2, weight.reshape(weight.shape[0], cin), bias, act, r2)
+    return y.reshape(b, h, w, weight.shape[0]).permute(0, 3, 1, 2)
 
 
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:

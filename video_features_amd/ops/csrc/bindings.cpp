@@ -44,8 +44,8 @@ void vfa_maxpool3d_same(const void*, void*, long long, int, int, int, int,
 void vfa_maxpool2d_same(const void*, void*, long long, int, int, int, int,
                         int, int, int, int, int, int, int, int, int,
                         hipStream_t);
-void vfa_linear_act(const void*, const void*, const void*, void*, int, int,
-                    int, int, hipStream_t);
+void vfa_linear_act(const void*, const void*, const void*, const void*,
+                    void*, int, int, int, int, hipStream_t);
 }
 
 namespace {
@@ -376,7 +376,8 @@ torch::Tensor maxpool2d_same(torch::Tensor x, std::vector<int64_t> kernel,
 }
 
 torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
-                         c10::optional<torch::Tensor> bias, int64_t act) {
+                         c10::optional<torch::Tensor> bias,
+                         c10::optional<torch::Tensor> res, int64_t act) {
   // x (M, K) bf16, w (N, K) bf16 (torch Linear layout) -> act(x@w^T+b)
   TORCH_CHECK(x.is_cuda() && x.dim() == 2 && x.is_contiguous());
   TORCH_CHECK(w.dim() == 2 && w.is_contiguous());
@@ -393,9 +394,17 @@ torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
     TORCH_CHECK(bc.numel() == n);
     bptr = bc.data_ptr();
   }
+  const void* rptr = nullptr;
+  if (res.has_value()) {
+    TORCH_CHECK(res->is_contiguous() &&
+                res->scalar_type() == torch::kBFloat16 &&
+                res->numel() == (long long)m * n,
+                "residual must be (M, N) bf16 contiguous");
+    rptr = res->data_ptr();
+  }
   auto out = torch::empty({m, n}, x.options());
-  vfa_linear_act(x.data_ptr(), w.data_ptr(), bptr, out.data_ptr(), m, n, kk,
-                 (int)act, current_stream());
+  vfa_linear_act(x.data_ptr(), w.data_ptr(), bptr, rptr, out.data_ptr(), m,
+                 n, kk, (int)act, current_stream());
   return out;
 }
 

@@ -95,6 +95,7 @@ __global__ __launch_bounds__(BIG ? 512 : 256)
 void linear_act_kernel(const __bf16* __restrict__ a,
                        const __bf16* __restrict__ w,
                        const __bf16* __restrict__ bias,
+                       const __bf16* __restrict__ res,
                        __bf16* __restrict__ c, int m, int n, int k,
                        int tiles_m, int tiles_n) {
   constexpr int BM = BIG ? 256 : 128, BN = BIG ? 256 : 128;
@@ -199,16 +200,18 @@ void linear_act_kernel(const __bf16* __restrict__ a,
       for (int r = 0; r < 4; ++r) {
         const int row = m0 + wm * (MI * 16) + i * 16 + hi4 * 4 + r;
         if (!tile_full && row >= m) continue;
-        c[(long long)row * n + col] =
-            (__bf16)act_f(acc[i][j][r] + bv, ACT);
+        float v = acc[i][j][r] + bv;
+        if (res) v += (float)res[(long long)row * n + col];
+        c[(long long)row * n + col] = (__bf16)act_f(v, ACT);
       }
     }
   }
 }
 
 template <int ACT, bool BIG>
-void launch_tile(const void* a, const void* w, const void* bias, void* c,
-                 int m, int n, int k, hipStream_t stream) {
+void launch_tile(const void* a, const void* w, const void* bias,
+                 const void* res, void* c, int m, int n, int k,
+                 hipStream_t stream) {
   constexpr int BM = BIG ? 256 : 128, BN = BIG ? 256 : 128;
   const int tiles_m = (m + BM - 1) / BM, tiles_n = (n + BN - 1) / BN;
   const dim3 grid(tiles_m * tiles_n);
@@ -217,26 +220,29 @@ void launch_tile(const void* a, const void* w, const void* bias, void* c,
   if (full)
     hipLaunchKernelGGL((linear_act_kernel<ACT, true, BIG>), grid,
                        dim3(BIG ? 512 : 256), lds, stream, (const __bf16*)a,
-                       (const __bf16*)w, (const __bf16*)bias, (__bf16*)c, m,
-                       n, k, tiles_m, tiles_n);
+                       (const __bf16*)w, (const __bf16*)bias,
+                       (const __bf16*)res, (__bf16*)c, m, n, k, tiles_m,
+                       tiles_n);
   else
     hipLaunchKernelGGL((linear_act_kernel<ACT, false, BIG>), grid,
                        dim3(BIG ? 512 : 256), lds, stream, (const __bf16*)a,
-                       (const __bf16*)w, (const __bf16*)bias, (__bf16*)c, m,
-                       n, k, tiles_m, tiles_n);
+                       (const __bf16*)w, (const __bf16*)bias,
+                       (const __bf16*)res, (__bf16*)c, m, n, k, tiles_m,
+                       tiles_n);
 }
 
 template <int ACT>
-void launch_linear(const void* a, const void* w, const void* bias, void* c,
-                   int m, int n, int k, hipStream_t stream) {
+void launch_linear(const void* a, const void* w, const void* bias,
+                   const void* res, void* c, int m, int n, int k,
+                   hipStream_t stream) {
   // BIG tiles when M tiles evenly (a half-empty 256-row tail tile and the
   // block-round quantization cost more than the smaller tile's overhead —
   // measured 339 vs 528 TF at M=9600) and the grid still fills the chip
   const long long tiles = (long long)((m + 255) / 256) * ((n + 255) / 256);
   if (n >= 256 && m % 256 == 0 && tiles >= 150)
-    launch_tile<ACT, true>(a, w, bias, c, m, n, k, stream);
+    launch_tile<ACT, true>(a, w, bias, res, c, m, n, k, stream);
   else
-    launch_tile<ACT, false>(a, w, bias, c, m, n, k, stream);
+    launch_tile<ACT, false>(a, w, bias, res, c, m, n, k, stream);
 }
 
 }  // namespace
@@ -244,13 +250,14 @@ void launch_linear(const void* a, const void* w, const void* bias, void* c,
 extern "C" {
 
 // act: 0 none, 1 relu, 2 quick_gelu, 3 gelu_tanh
-void vfa_linear_act(const void* a, const void* w, const void* bias, void* c,
-                    int m, int n, int k, int act, hipStream_t stream) {
+void vfa_linear_act(const void* a, const void* w, const void* bias,
+                    const void* res, void* c, int m, int n, int k, int act,
+                    hipStream_t stream) {
   switch (act) {
-    case 0: launch_linear<0>(a, w, bias, c, m, n, k, stream); break;
-    case 1: launch_linear<1>(a, w, bias, c, m, n, k, stream); break;
-    case 2: launch_linear<2>(a, w, bias, c, m, n, k, stream); break;
-    case 3: launch_linear<3>(a, w, bias, c, m, n, k, stream); break;
+    case 0: launch_linear<0>(a, w, bias, res, c, m, n, k, stream); break;
+    case 1: launch_linear<1>(a, w, bias, res, c, m, n, k, stream); break;
+    case 2: launch_linear<2>(a, w, bias, res, c, m, n, k, stream); break;
+    case 3: launch_linear<3>(a, w, bias, res, c, m, n, k, stream); break;
   }
 }
 
