@@ -177,12 +177,14 @@ void linear_act_kernel(const __bf16* __restrict__ a,
         afr[i] = *reinterpret_cast<const bf16x8*>(
             sA(cur) + swz(arow * 128 + kk * 64 + hi4 * 16));
       }
+      __builtin_amdgcn_s_setprio(1);   // keep the MFMA cluster issuing
 #pragma unroll
       for (int i = 0; i < MI; ++i)
 #pragma unroll
         for (int j = 0; j < NJ; ++j)
           acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afr[i], bfr[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
     }
     // no trailing barrier: the next iteration's top __syncthreads() both
     // drains the in-flight glds and orders reads before buffer reuse
