@@ -67,3 +67,19 @@ def test_maxpool3d_same_matches_i3d_geometry(dev):
     assert out.shape == (1, 4, 64, 112, 112)
     out = ops.maxpool3d_same(x, (3, 3, 3), (2, 2, 2))
     assert out.shape == (1, 4, 32, 112, 112)
+
+
+@pytest.mark.parametrize('kt,st,p0,t', [(3, 1, 1, 8), (3, 2, 1, 16),
+                                        (3, 2, 1, 15)])
+def test_temporal_merge_fused(dev, kt, st, p0, t):
+    """Fused temporal tap merge vs the strided-add torch composition."""
+    from video_features_amd.models import _flat3d
+    torch.manual_seed(0)
+    b, o, h, w = 2, 24, 5, 7
+    y = torch.randn(b * t, kt * o, h, w, device=dev) \
+        .contiguous(memory_format=torch.channels_last)
+    out = _flat3d.temporal_merge(y, b, kt, st, p0)      # fused on GPU
+    ref = _flat3d.temporal_merge(y.cpu().float(), b, kt, st, p0)
+    assert out.shape == ref.shape
+    assert torch.allclose(out.cpu().float(), ref, atol=1e-5), \
+        (out.cpu().float() - ref).abs().max().item()

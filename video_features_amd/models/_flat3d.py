@@ -17,6 +17,8 @@ from __future__ import annotations
 
 import torch
 
+from .. import ops
+
 
 def flatten_time(x: torch.Tensor) -> torch.Tensor:
     """(B, C, T, H, W) → (B*T, C, H, W) channels_last, one copy."""
@@ -51,6 +53,9 @@ def temporal_merge(y: torch.Tensor, b: int, kt: int, st: int = 1,
     to = (t + 2 * p0 - kt) // st + 1
     if bias_tap is None:
         bias_tap = kt // 2
+    fused = ops.temporal_merge_fused(y, b, kt, st, p0)
+    if fused is not None:
+        return fused
     out = cl_empty(b * to, o, h, w, y)
     y5 = y.view(b, t, ckt, h, w)
     o5 = out.view(b, to, o, h, w)

@@ -377,6 +377,17 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
     return y
 
 
+def temporal_merge_fused(y: torch.Tensor, b: int, kt: int, st: int,
+                         p0: int) -> Optional[torch.Tensor]:
+    """One-kernel temporal tap merge for the flattened-time conv3d
+    decomposition (see models/_flat3d.py).  Returns None when the HIP path
+    does not apply (caller falls back to the strided-add composition)."""
+    if (_use_hip(y)
+            and y.is_contiguous(memory_format=torch.channels_last)):
+        return _ext.temporal_merge(y, b, kt, st, p0)
+    return None
+
+
 def conv1x1_act(x: torch.Tensor, weight: torch.Tensor,
                 bias: Optional[torch.Tensor] = None, act: str = 'none',
                 res: Optional[torch.Tensor] = None) -> torch.Tensor:

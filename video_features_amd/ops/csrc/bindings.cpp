@@ -46,6 +46,8 @@ void vfa_maxpool2d_same(const void*, void*, long long, int, int, int, int,
                         hipStream_t);
 void vfa_linear_act(const void*, const void*, const void*, const void*,
                     void*, int, int, int, int, hipStream_t);
+void vfa_temporal_merge(const void*, void*, int, int, int, int, int, int,
+                        int, long long, int, hipStream_t);
 }
 
 namespace {
@@ -408,6 +410,23 @@ torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
   return out;
 }
 
+torch::Tensor temporal_merge(torch::Tensor y, int64_t b, int64_t kt,
+                             int64_t st, int64_t p0) {
+  // y (B*T, kt*O, H, W) channels_last -> (B*T', O, H, W) channels_last
+  TORCH_CHECK(y.is_cuda() && y.dim() == 4);
+  TORCH_CHECK(cl_contig(y), "channels_last expected");
+  const int bt = (int)y.size(0), cin = (int)y.size(1);
+  const int h = (int)y.size(2), w = (int)y.size(3);
+  const int o = cin / (int)kt, t = bt / (int)b;
+  const int to = (int)((t + 2 * p0 - kt) / st + 1);
+  auto out = torch::empty({(long)b * to, h, w, o}, y.options())
+                 .permute({0, 3, 1, 2});
+  vfa_temporal_merge(y.data_ptr(), out.data_ptr(), (int)b, t, to, (int)kt,
+                     (int)st, (int)p0, o, (long long)h * w, dtype_tag(y),
+                     current_stream());
+  return out;
+}
+
 }  // namespace
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
@@ -430,5 +449,6 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3d_same", &maxpool3d_same);
   m.def("maxpool2d_same", &maxpool2d_same);
   m.def("linear_act", &linear_act);
+  m.def("temporal_merge", &temporal_merge);
   m.attr("gfx_arch") = "gfx950";
 }
