@@ -23,6 +23,10 @@ SHAPES = [
     (38400, 3072, 768, 'quick_gelu', 'CLIP fc1 (fb768)'),
     (4096, 4096, 12288, 'relu', 'VGGish fc1'),
     (200704, 256, 1920, 'none', 'RAFT GRU-zr-as-GEMM'),
+    (4096, 4096, 4096, 'none', 'ladder 4096^3'),
+    (8192, 8192, 8192, 'none', 'ladder 8192^3'),
+    (19200, 3072, 768, 'quick_gelu', 'CLIP fc1 (fb384)'),
+    (19200, 2304, 768, 'none', 'CLIP qkv (fb384)'),
 ]
 
 
