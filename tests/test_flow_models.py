@@ -121,3 +121,18 @@ def test_raft_checkpoint_compat_separate_zr():
     gru2.load_state_dict(legacy)
     for k, v in gru2.state_dict().items():
         assert torch.equal(v, sd[k]), k
+
+
+def test_show_pred_saves_flow_visualization(tmp_path):
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.pwc import ExtractPWC
+    from video_features_amd.io.y4m import write_y4m
+    import os
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=5, h=64, w=64), fps=25.0)
+    cfg = Config(feature_type='pwc', video_paths=[vid], cpu=True,
+                 show_pred=True, tmp_path=str(tmp_path / 'tmp'))
+    out = ExtractPWC(cfg, external_call=True)(torch.arange(1))[0]
+    assert out['pwc'].shape[0] == 4
+    ppms = [f for f in os.listdir(tmp_path / 'tmp') if f.endswith('.ppm')]
+    assert ppms, 'flow visualization not saved'

@@ -54,9 +54,17 @@ class ExtractPWC(BaseExtractor):
             flow = model(batch[:-1], batch[1:])
             flows.append(flow.float().cpu())
             if self.show_pred:
+                # headless flow visualization (see raft extractor note)
+                from ..utils.flow_viz import flow_to_image, save_ppm
+                import os as _os
                 mag = flow.norm(dim=1)
                 print(f'flow frames {start}-{stop - 1}: '
                       f'|flow| mean {mag.mean():.3f} max {mag.max():.3f}')
+                _os.makedirs(self.tmp_path, exist_ok=True)
+                img = flow_to_image(
+                    flow[0].float().cpu().numpy().transpose(1, 2, 0))
+                save_ppm(_os.path.join(
+                    self.tmp_path, f'flow_vis_{start:05d}.ppm'), img)
             start = stop - 1
         features = torch.cat(flows).numpy() if flows else np.zeros((0, 2, 0, 0))
         return {

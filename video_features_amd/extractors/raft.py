@@ -62,9 +62,20 @@ class ExtractRAFT(BaseExtractor):
             flow = model(im1, im2, test_mode=True)
             flows.append(padder.unpad(flow).float().cpu())
             if self.show_pred:
+                # the reference pops a cv2 window of the Middlebury flow
+                # visualization (reference extract_raft.py:165-178); this
+                # environment is headless, so save the image + print stats
+                from ..utils.flow_viz import flow_to_image, save_ppm
+                import os as _os
                 mag = flow.norm(dim=1)
                 print(f'flow frames {start}-{stop - 1}: '
                       f'|flow| mean {mag.mean():.3f} max {mag.max():.3f}')
+                _os.makedirs(self.tmp_path, exist_ok=True)
+                img = flow_to_image(
+                    padder.unpad(flow)[0].float().cpu().numpy()
+                    .transpose(1, 2, 0))
+                save_ppm(_os.path.join(
+                    self.tmp_path, f'flow_vis_{start:05d}.ppm'), img)
             start = stop - 1   # carry the last frame over for continuity
         features = torch.cat(flows).numpy() if flows else np.zeros((0, 2, 0, 0))
         return {

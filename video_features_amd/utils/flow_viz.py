@@ -56,3 +56,12 @@ def flow_to_image(flow: np.ndarray, clip_max: float = None) -> np.ndarray:
         v = np.clip(v, -clip_max, clip_max)
     rad_max = max(np.sqrt(u ** 2 + v ** 2).max(), 1e-5)
     return flow_uv_to_colors(u / rad_max, v / rad_max)
+
+
+def save_ppm(path: str, rgb: 'np.ndarray') -> None:
+    """Write an (H, W, 3) uint8 image as binary PPM (no PIL dependency)."""
+    import numpy as np
+    rgb = np.ascontiguousarray(rgb, dtype=np.uint8)
+    with open(path, 'wb') as f:
+        f.write(b'P6\n%d %d\n255\n' % (rgb.shape[1], rgb.shape[0]))
+        f.write(rgb.tobytes())
