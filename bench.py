@@ -150,17 +150,21 @@ def bench_i3d_raft(args, device, dtype, rank, world):
     from video_features_amd import transforms as T
     from video_features_amd.models.i3d import I3D
     from video_features_amd.models.raft import RAFT
+    from video_features_amd.models.pwc import PWCNet
     torch.manual_seed(0)
     from video_features_amd.utils.fold_bn import fold_batchnorms
     i3d_rgb = I3D(modality='rgb').to(device, dtype).eval()
     i3d_flow = I3D(modality='flow').to(device, dtype).eval()
-    raft = RAFT(iters=args.raft_iters).to(device, dtype).eval()
+    if args.flow == 'pwc':
+        raft = PWCNet().to(device, dtype).eval()
+    else:
+        raft = RAFT(iters=args.raft_iters).to(device, dtype).eval()
     # inference-only: fold BN affine maps into the convs (see fold_bn.py)
     fold_batchnorms(i3d_rgb)
     fold_batchnorms(i3d_flow)
     fold_batchnorms(raft)
     nhwc = args.layout == 'nhwc' and device.type == 'cuda'
-    if nhwc:
+    if nhwc and hasattr(raft, 'use_channels_last'):
         raft = raft.use_channels_last()
     cl3d = args.i3d_cl3d and device.type == 'cuda'
     if cl3d:
@@ -187,7 +191,10 @@ def bench_i3d_raft(args, device, dtype, rank, world):
         x = frames_u8_dev.permute(0, 1, 4, 2, 3).to(dtype)
         i1 = x[:, :-1].reshape(-1, 3, 224, 224)
         i2 = x[:, 1:].reshape(-1, 3, 224, 224)
-        flow = raft(i1, i2, test_mode=True)          # (clips*64, 2, H, W)
+        if args.flow == 'pwc':
+            flow = raft(i1, i2)                      # (clips*64, 2, H, W)
+        else:
+            flow = raft(i1, i2, test_mode=True)
         rgb_in = T.scale_to_pm1(x[:, :-1]).transpose(1, 2)
         flow_in = T.i3d_flow_preprocess(flow, 224) \
             .reshape(clips, stack, 2, 224, 224).transpose(1, 2)
@@ -242,11 +249,12 @@ def bench_i3d_raft(args, device, dtype, rank, world):
     dt = max_over_ranks(dt, device, world)
     total_clips = clips * world * args.steps
     return {
-        'metric': 'clips/sec I3D rgb+flow (RAFT)',
+        'metric': f'clips/sec I3D rgb+flow ({args.flow.upper()})',
         'value': total_clips / dt,
         'unit': 'clips/sec',
         'ms_per_step': dt / args.steps * 1000.0,
-        'config': {'model': 'I3D+RAFT', 'global_batch': clips * world,
+        'config': {'model': f'I3D+{args.flow.upper()}',
+                   'global_batch': clips * world,
                    'seq_len': stack, 'resolution': 224,
                    'raft_iters': args.raft_iters, 'layout': args.layout,
                    'parallelism': f'dp{world}'},
@@ -398,6 +406,8 @@ def main():
                    help='CLIP: frames per forward chunk')
     p.add_argument('--clips-per-step', type=int, default=2)
     p.add_argument('--raft-iters', type=int, default=20)
+    p.add_argument('--flow', choices=['raft', 'pwc'], default='raft',
+                   help='flow net for the i3d_raft bench')
     p.add_argument('--layout', choices=['nhwc', 'nchw'], default='nhwc',
                    help='RAFT conv layout on GPU (nhwc = channels_last)')
     p.add_argument('--i3d-cl3d', action='store_true',
