@@ -100,3 +100,29 @@ def test_print_sink_smoke(capsys):
                          'print', False, 'f')
     out = capsys.readouterr().out
     assert 'shape (2, 3)' in out
+
+
+def test_main_cli_end_to_end(tmp_path):
+    """python main.py ... on a real file → .npy outputs (the reference's
+    primary entry, reference main.py:93-149)."""
+    import subprocess
+    import sys
+    import os
+    import numpy as np
+    from tests.conftest import synthetic_frames
+    from video_features_amd.io.y4m import write_y4m
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=20, h=64, w=64), fps=25.0)
+    r = subprocess.run(
+        [sys.executable, 'main.py', '--feature_type', 'resnet18', '--cpu',
+         '--video_paths', vid, '--on_extraction', 'save_numpy',
+         '--output_path', str(tmp_path / 'out'),
+         '--tmp_path', str(tmp_path / 'tmp'), '--batch_size', '8'],
+        capture_output=True, text=True, cwd=root, timeout=600)
+    assert r.returncode == 0, r.stderr[-2000:]
+    out = tmp_path / 'out' / 'resnet18'
+    arrs = {f: np.load(out / f) for f in os.listdir(out)}
+    feats = [a for f, a in arrs.items() if 'resnet18' in f
+             and 'fps' not in f and 'timestamps' not in f]
+    assert feats and feats[0].shape == (20, 512)
