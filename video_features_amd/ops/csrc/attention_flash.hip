@@ -96,7 +96,19 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
   const long long row_stride = 3LL * h_total * D;
   const __bf16* q_g = qkv + ((long long)bi * n + qbase) * row_stride +
                       (long long)hi * D;                    // +0 for q
+  const int n_kv0 = (n + KVBLK - 1) / KVBLK;
   stage_rows(q_g, QBLK, row_stride, s_q, max(0, n - qbase));
+  if (n_kv0 == 1) {
+    // ViT-length sequences (N <= 64): one KV tile — stage Q, K, V together
+    // so all staging loads are in flight at once behind a SINGLE barrier
+    // (PMC: this kernel was 66% wave-parked with the 3-barrier schedule)
+    const __bf16* k_g0 = qkv + (long long)bi * n * row_stride +
+                         ((long long)1 * h_total + hi) * D;
+    const __bf16* v_g0 = qkv + (long long)bi * n * row_stride +
+                         ((long long)2 * h_total + hi) * D;
+    stage_rows(k_g0, KVBLK, row_stride, s_k, n);
+    stage_vt(v_g0, row_stride, s_vt, n);
+  }
   __syncthreads();
 
   // per-wave state: 16 query rows [wave*16, wave*16+16)
@@ -107,18 +119,20 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
 #pragma unroll
   for (int r = 0; r < 4; ++r) { m_run[r] = -1e30f; l_run[r] = 0.f; }
 
-  const int n_kv = (n + KVBLK - 1) / KVBLK;
+  const int n_kv = n_kv0;
   for (int kv = 0; kv < n_kv; ++kv) {
     const int kvbase = kv * KVBLK;
     const int valid = min(KVBLK, n - kvbase);
-    __syncthreads();
-    const __bf16* k_g = qkv + ((long long)bi * n + kvbase) * row_stride +
-                        ((long long)1 * h_total + hi) * D;
-    const __bf16* v_g = qkv + ((long long)bi * n + kvbase) * row_stride +
-                        ((long long)2 * h_total + hi) * D;
-    stage_rows(k_g, KVBLK, row_stride, s_k, valid);
-    stage_vt(v_g, row_stride, s_vt, valid);
-    __syncthreads();
+    if (n_kv > 1) {   // single-tile case staged K/V with Q above
+      __syncthreads();
+      const __bf16* k_g = qkv + ((long long)bi * n + kvbase) * row_stride +
+                          ((long long)1 * h_total + hi) * D;
+      const __bf16* v_g = qkv + ((long long)bi * n + kvbase) * row_stride +
+                          ((long long)2 * h_total + hi) * D;
+      stage_rows(k_g, KVBLK, row_stride, s_k, valid);
+      stage_vt(v_g, row_stride, s_vt, valid);
+      __syncthreads();
+    }
 
     // ---- S = Q K^T for this wave's 16 rows x 64 keys
     f32x4 s_frag[4];
