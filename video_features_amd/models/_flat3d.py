@@ -20,6 +20,26 @@ import torch
 from .. import ops
 
 
+def cached_cl_weight(mod, name: str, src: torch.Tensor, build):
+    """Per-module cache of transformed conv weights for the flat path.
+
+    The flat decomposition slices/permutes the stored Conv3d weights; the
+    resulting views are NOT channels_last-contiguous, and F.conv2d with a
+    non-CL weight silently drops MIOpen's NHWC igemm path (measured: I3D at
+    17 TF/s effective vs RAFT's 911 with module convs).  Build once, store
+    CL-contiguous, invalidate on weight version/dtype/device change."""
+    key = (src._version, src.dtype, src.device, src.data_ptr())
+    cache = mod.__dict__.setdefault('_flat_w_cache', {})
+    ent = cache.get(name)
+    if ent is None or ent[0] != key:
+        val = build()
+        if isinstance(val, torch.Tensor) and val.dim() == 4:
+            val = val.contiguous(memory_format=torch.channels_last)
+        cache[name] = (key, val)
+        return val
+    return ent[1]
+
+
 def flatten_time(x: torch.Tensor) -> torch.Tensor:
     """(B, C, T, H, W) → (B*T, C, H, W) channels_last, one copy."""
     b, c, t, h, w = x.shape

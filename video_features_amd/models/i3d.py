@@ -47,7 +47,7 @@ def _triple(v) -> Tuple[int, int, int]:
 # See models/_flat3d.py: the backbone below the stem runs on (B*T, C, H, W)
 # channels_last tensors (no conv3d, no Im3d2Col).
 from ._flat3d import (flatten_time, unflatten_time, cl_empty,
-                      temporal_merge, temporal_max)
+                      temporal_merge, temporal_max, cached_cl_weight)
 
 
 def temporal_merge3(y: torch.Tensor, b: int) -> torch.Tensor:
@@ -93,16 +93,22 @@ class Unit3D(nn.Module):
         w5 = self.conv.weight
         bias = self.conv.bias
         if kt == 1:
-            x = F.conv2d(xf, w5[:, :, 0], bias,
+            w2 = cached_cl_weight(self, 'w2', w5, lambda: w5[:, :, 0])
+            x = F.conv2d(xf, w2, bias,
                          padding=(self.kernel[1] // 2, self.kernel[2] // 2))
         else:  # 3x3x3, stride 1
             o = w5.shape[0]
-            wcat = w5.permute(2, 0, 1, 3, 4).reshape(
-                3 * o, w5.shape[1], self.kernel[1], self.kernel[2])
+            wcat = cached_cl_weight(
+                self, 'wcat', w5,
+                lambda: w5.permute(2, 0, 1, 3, 4).reshape(
+                    3 * o, w5.shape[1], self.kernel[1], self.kernel[2]))
             if bias is not None:
-                bcat = torch.zeros(3 * o, device=bias.device,
-                                   dtype=bias.dtype)
-                bcat[o:2 * o] = bias
+                def mk_bcat():
+                    bc = torch.zeros(3 * o, device=bias.device,
+                                     dtype=bias.dtype)
+                    bc[o:2 * o] = bias
+                    return bc
+                bcat = cached_cl_weight(self, 'bcat', bias, mk_bcat)
             else:
                 bcat = None
             y = F.conv2d(xf, wcat, bcat,

@@ -16,7 +16,8 @@ import torch
 import torch.nn.functional as F
 from torch import nn
 
-from ._flat3d import (flatten_time, temporal_merge, temporal_select)
+from ._flat3d import (flatten_time, temporal_merge, temporal_select,
+                      cached_cl_weight)
 
 
 def _flat_bn_relu(x, bn, relu: bool):
@@ -51,16 +52,22 @@ class Conv2Plus1D(nn.Module):
         spatial (1,3,3) conv = conv2d; temporal (3,1,1) conv = merged
         3-tap 1x1 conv2d + shifted (strided) temporal add."""
         ss = self.spatial.stride[1]
-        y = F.conv2d(xf, self.spatial.weight[:, :, 0], self.spatial.bias,
-                     stride=ss, padding=1)
+        sw = cached_cl_weight(self, 'sw', self.spatial.weight,
+                              lambda: self.spatial.weight[:, :, 0])
+        y = F.conv2d(xf, sw, self.spatial.bias, stride=ss, padding=1)
         y = _flat_bn_relu(y, self.bn, True)
         w = self.temporal.weight                   # (O, M, 3, 1, 1)
         o = w.shape[0]
-        wcat = w.permute(2, 0, 1, 3, 4).reshape(3 * o, w.shape[1], 1, 1)
+        wcat = cached_cl_weight(
+            self, 'wcat', w,
+            lambda: w.permute(2, 0, 1, 3, 4).reshape(3 * o, w.shape[1], 1, 1))
         bias = self.temporal.bias
         if bias is not None:
-            bcat = torch.zeros(3 * o, device=bias.device, dtype=bias.dtype)
-            bcat[o:2 * o] = bias
+            def mk_bcat():
+                bc = torch.zeros(3 * o, device=bias.device, dtype=bias.dtype)
+                bc[o:2 * o] = bias
+                return bc
+            bcat = cached_cl_weight(self, 'bcat', bias, mk_bcat)
         else:
             bcat = None
         y = F.conv2d(y, wcat, bcat)
@@ -99,7 +106,9 @@ class R21DBlock(nn.Module):
             st = conv.stride[0]
             # 1x1x1 stride-s conv: temporal subsample, then conv2d stride s
             identity = temporal_select(xf, b, st)
-            identity = F.conv2d(identity, conv.weight[:, :, 0], conv.bias,
+            dw = cached_cl_weight(self, 'dw', conv.weight,
+                                  lambda: conv.weight[:, :, 0])
+            identity = F.conv2d(identity, dw, conv.bias,
                                 stride=conv.stride[1])
             identity = _flat_bn_relu(identity, bn, False)
         out = self.conv1.forward_flat(xf, b)
@@ -135,16 +144,22 @@ class R2Plus1D18(nn.Module):
         # stem: (1,7,7)/s(1,2,2) conv2d -> BN+ReLU -> 3-tap temporal 1x1
         sc0, sbn0, sc1, sbn1 = (self.stem[0], self.stem[1], self.stem[3],
                                 self.stem[4])
-        xf = F.conv2d(xf, sc0.weight[:, :, 0], sc0.bias, stride=2, padding=3)
+        s0w = cached_cl_weight(self, 's0w', sc0.weight,
+                               lambda: sc0.weight[:, :, 0])
+        xf = F.conv2d(xf, s0w, sc0.bias, stride=2, padding=3)
         xf = _flat_bn_relu(xf, sbn0, True)
         o = sc1.weight.shape[0]
-        wcat = sc1.weight.permute(2, 0, 1, 3, 4).reshape(3 * o,
-                                                         sc1.weight.shape[1],
-                                                         1, 1)
+        wcat = cached_cl_weight(
+            self, 's1w', sc1.weight,
+            lambda: sc1.weight.permute(2, 0, 1, 3, 4).reshape(
+                3 * o, sc1.weight.shape[1], 1, 1))
         if sc1.bias is not None:
-            bcat = torch.zeros(3 * o, device=sc1.bias.device,
-                               dtype=sc1.bias.dtype)
-            bcat[o:2 * o] = sc1.bias
+            def mk_bcat():
+                bc = torch.zeros(3 * o, device=sc1.bias.device,
+                                 dtype=sc1.bias.dtype)
+                bc[o:2 * o] = sc1.bias
+                return bc
+            bcat = cached_cl_weight(self, 's1b', sc1.bias, mk_bcat)
         else:
             bcat = None
         xf = temporal_merge(F.conv2d(xf, wcat, bcat), b, kt=3, st=1, p0=1)
