@@ -94,6 +94,16 @@ class Unit3D(nn.Module):
         bias = self.conv.bias
         if kt == 1:
             w2 = cached_cl_weight(self, 'w2', w5, lambda: w5[:, :, 0])
+            if (self.kernel[1] == 1 and xf.is_cuda
+                    and xf.dtype == torch.bfloat16
+                    and not isinstance(self.bn, nn.BatchNorm3d)
+                    and w5.shape[1] % 8 == 0 and ops.hip_available()
+                    and xf.is_contiguous(memory_format=torch.channels_last)):
+                # folded 1x1x1 conv (+bias+ReLU) as ONE fused MFMA GEMM on
+                # the CL view: removes MIOpen's SubTensor zero-fill, the
+                # bias pass, and the separate ReLU round trip
+                return ops.conv1x1_act(
+                    xf, w2, bias, 'relu' if self.activation else 'none')
             x = F.conv2d(xf, w2, bias,
                          padding=(self.kernel[1] // 2, self.kernel[2] // 2))
         else:  # 3x3x3, stride 1
