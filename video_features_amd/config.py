@@ -12,7 +12,7 @@ Defaults mirror the reference CLI defaults (reference main.py:94-137).
 from __future__ import annotations
 
 import dataclasses
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from typing import Any, List, Optional, Sequence
 
 FEATURE_TYPES = [

@@ -7,10 +7,9 @@ numpy for decode, scipy.signal.resample_poly for resampling.
 """
 from __future__ import annotations
 
-import os
 import wave
 from pathlib import Path
-from typing import Optional, Tuple
+from typing import Tuple
 
 import numpy as np
 

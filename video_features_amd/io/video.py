@@ -24,7 +24,7 @@ from __future__ import annotations
 import os
 import re
 from pathlib import Path
-from typing import List, Optional, Sequence
+from typing import Optional
 
 import numpy as np
 

@@ -8,9 +8,8 @@ as the interchange format between an external ffmpeg and this framework.
 """
 from __future__ import annotations
 
-import io
 import os
-from typing import List, Tuple
+from typing import Tuple
 
 import numpy as np
 

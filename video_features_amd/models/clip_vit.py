@@ -13,7 +13,6 @@ rocBLAS/hipBLASLt via ``torch.nn.Linear``.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 
 import torch

@@ -46,8 +46,8 @@ def _triple(v) -> Tuple[int, int, int]:
 # ----------------------------------------------------- flattened-time path
 # See models/_flat3d.py: the backbone below the stem runs on (B*T, C, H, W)
 # channels_last tensors (no conv3d, no Im3d2Col).
-from ._flat3d import (flatten_time, unflatten_time, cl_empty,
-                      temporal_merge, temporal_max, cached_cl_weight)
+from ._flat3d import (flatten_time, unflatten_time, temporal_merge,
+                      temporal_max, cached_cl_weight)
 
 
 def temporal_merge3(y: torch.Tensor, b: int) -> torch.Tensor:

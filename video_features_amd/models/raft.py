@@ -24,7 +24,7 @@ MI355X mapping (all decisions measured with rocprofv3, profiles/):
 """
 from __future__ import annotations
 
-from typing import List, Tuple
+from typing import List
 
 import torch
 import torch.nn.functional as F

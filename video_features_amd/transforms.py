@@ -12,7 +12,7 @@ for 3D nets are ``(C, T, H, W)``.
 from __future__ import annotations
 
 import math
-from typing import Optional, Sequence, Tuple
+from typing import Sequence
 
 import torch
 import torch.nn.functional as F
