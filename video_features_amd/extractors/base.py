@@ -64,6 +64,19 @@ class BaseExtractor(torch.nn.Module):
                 video_path) -> Dict[str, np.ndarray]:
         raise NotImplementedError
 
+    @staticmethod
+    def load_weights(model: torch.nn.Module, path: str) -> None:
+        """Load a state-dict file, accepting the published reference
+        checkpoint schemes (OpenAI CLIP / reference i3d / torchvision
+        r2plus1d / torchvggish / DataParallel 'module.' prefixes) via
+        utils.convert_checkpoints.convert_auto."""
+        from ..utils.convert_checkpoints import convert_auto
+        sd = torch.load(path, map_location='cpu', weights_only=True)
+        if isinstance(sd, dict) and 'state_dict' in sd \
+                and isinstance(sd['state_dict'], dict):
+            sd = sd['state_dict']
+        model.load_state_dict(convert_auto(sd))
+
     def _prof(self, stage: str):
         from ..runtime.profiler import StageProfiler
         if self.prof is None:

@@ -29,9 +29,7 @@ class ExtractCLIP(BaseExtractor):
     def build_models(self, device: torch.device, dtype: torch.dtype):
         model = build_clip_vit(self.feature_type)
         if self.cfg.weights_path:
-            sd = torch.load(self.cfg.weights_path, map_location='cpu',
-                            weights_only=True)
-            model.load_state_dict(sd)
+            self.load_weights(model, self.cfg.weights_path)
         model = model.to(device=device, dtype=dtype).eval()
         return model
 

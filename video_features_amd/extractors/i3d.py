@@ -57,9 +57,7 @@ class ExtractI3D(BaseExtractor):
                 p = p.with_suffix('.pth')
         if not p.exists():
             return
-        sd = torch.load(str(p), map_location='cpu', weights_only=True)
-        sd = {k.removeprefix('module.'): v for k, v in sd.items()}
-        model.load_state_dict(sd)
+        self.load_weights(model, str(p))
 
     def build_models(self, device: torch.device, dtype: torch.dtype):
         models = {}

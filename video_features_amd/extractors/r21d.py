@@ -31,9 +31,7 @@ class ExtractR21D(BaseExtractor):
     def build_models(self, device: torch.device, dtype: torch.dtype):
         model = R2Plus1D18()
         if self.cfg.weights_path:
-            sd = torch.load(self.cfg.weights_path, map_location='cpu',
-                            weights_only=True)
-            model.load_state_dict(sd)
+            self.load_weights(model, self.cfg.weights_path)
         return model.to(device=device, dtype=dtype).eval()
 
     def extract(self, device: torch.device, model,

@@ -30,12 +30,7 @@ class ExtractRAFT(BaseExtractor):
     def build_models(self, device: torch.device, dtype: torch.dtype):
         model = RAFT()
         if self.cfg.weights_path:
-            sd = torch.load(self.cfg.weights_path, map_location='cpu',
-                            weights_only=True)
-            # the published RAFT ckpts are DataParallel-wrapped (reference
-            # extract_raft.py:58-60 wraps the model instead; we strip)
-            sd = {k.removeprefix('module.'): v for k, v in sd.items()}
-            model.load_state_dict(sd)
+            self.load_weights(model, self.cfg.weights_path)
         return model.to(device=device, dtype=dtype).eval()
 
     def _prep(self, frames_u8: torch.Tensor) -> torch.Tensor:
