@@ -42,19 +42,23 @@ __device__ __forceinline__ float act_f(float x, int kind) {
 
 // stage a (ROWS x 64) bf16 tile into an LDS image with the st_16x32
 // swizzle via glds; each wave covers (ROWS*128/1024)/WAVES subtiles.
+// PIECES > 1 splits the wave's subtiles into issue groups so the glds for
+// the next K-tile can be spread across the current tile's MFMA quadrants
+// (piece = which group to issue; -1 = all).
 template <int ROWS, int WAVES>
-__device__ __forceinline__ void stage_glds(const __bf16* __restrict__ g,
-                                           long long row_stride_elems,
-                                           char* lds_base, int wave,
-                                           int lane) {
+__device__ __forceinline__ void stage_glds_piece(
+    const __bf16* __restrict__ g, long long row_stride_elems, char* lds_base,
+    int wave, int lane, int piece, int npieces) {
   constexpr int NSUB = ROWS * 128 / 1024;
+  constexpr int PER_WAVE = NSUB / WAVES;
   const int off = lane * 16;                   // linear LDS offset in subtile
   const int off_log = off ^ (((off >> 9) & 1) << 5);
   const int r_in = off_log >> 7;               // row within subtile
   const int b_in = off_log & 127;              // byte within 128-B row
 #pragma unroll
-  for (int i = 0; i < NSUB / WAVES; ++i) {
-    const int sub = wave * (NSUB / WAVES) + i;
+  for (int i = 0; i < PER_WAVE; ++i) {
+    if (piece >= 0 && (i * npieces) / PER_WAVE != piece) continue;
+    const int sub = wave * PER_WAVE + i;
     const __bf16* src = g + (long long)(sub * 8 + r_in) * row_stride_elems;
     // LDS destination is wave-uniform base + lane*16 (hardware-added);
     // the per-lane *global* address carries the swizzle
@@ -63,6 +67,15 @@ __device__ __forceinline__ void stage_glds(const __bf16* __restrict__ g,
             reinterpret_cast<const char*>(src) + b_in),
         reinterpret_cast<unsigned int*>(lds_base + sub * 1024), 16, 0, 0);
   }
+}
+
+template <int ROWS, int WAVES>
+__device__ __forceinline__ void stage_glds(const __bf16* __restrict__ g,
+                                           long long row_stride_elems,
+                                           char* lds_base, int wave,
+                                           int lane) {
+  stage_glds_piece<ROWS, WAVES>(g, row_stride_elems, lds_base, wave, lane,
+                                -1, 1);
 }
 
 // bounds-checked fallback staging (tails): same LDS image, zero fill.
@@ -160,8 +173,13 @@ void linear_act_kernel(const __bf16* __restrict__ a,
   for (int kt = 0; kt < nk; ++kt) {
     __syncthreads();                           // tile kt resident
     const int cur = kt & 1;
-    if (kt + 1 < nk) stage(1 - cur, kt + 1);
-    // compute on tile kt
+    const long long ko_nxt = (long long)(kt + 1) * BK;
+    const bool glds_nxt =
+        (kt + 1 < nk) && tile_full && (FULL || kt + 2 < nk);
+    if ((kt + 1 < nk) && !glds_nxt) stage(1 - cur, kt + 1);
+    // compute on tile kt; the next tile's glds issue is spread across the
+    // two MFMA half-steps so the memory pipe overlaps the math instead of
+    // bursting at the barrier (PMC: 39% of wave cycles were parked)
 #pragma unroll
     for (int kk = 0; kk < 2; ++kk) {           // two 32-deep MFMA steps
       bf16x8 afr[MI], bfr[NJ];
@@ -176,6 +194,12 @@ void linear_act_kernel(const __bf16* __restrict__ a,
         const int arow = wm * (MI * 16) + i * 16 + lo;
         afr[i] = *reinterpret_cast<const bf16x8*>(
             sA(cur) + swz(arow * 128 + kk * 64 + hi4 * 16));
+      }
+      if (glds_nxt) {
+        stage_glds_piece<BM, WAVES>(a + (long long)m0 * k + ko_nxt, k,
+                                    sA(1 - cur), wave, lane, kk, 2);
+        stage_glds_piece<BN, WAVES>(w + (long long)n0 * k + ko_nxt, k,
+                                    sB(1 - cur), wave, lane, kk, 2);
       }
       __builtin_amdgcn_s_setprio(1);   // keep the MFMA cluster issuing
 #pragma unroll
