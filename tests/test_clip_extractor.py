@@ -74,3 +74,27 @@ def test_error_isolation(tmp_path, y4m_video, capsys):
     feats_list = ex(torch.arange(2))
     assert len(feats_list) == 1      # good video still extracted
     assert 'Extraction failed' in capsys.readouterr().out
+
+
+def test_clip_golden_regression(tmp_path):
+    """Golden-feature regression: fixed seed-0 weights + fixed
+    synthetic video must reproduce the recorded features.  Guards silent
+    numerics drift in the preprocess/sampler/model chain."""
+    import json
+    import os
+    from tests.conftest import synthetic_frames
+    from video_features_amd.io.y4m import write_y4m
+    gold = json.load(open(os.path.join(os.path.dirname(__file__),
+                                       'golden_clip.json')))
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=24, h=64, w=96, seed=3), fps=25.0)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[vid], cpu=True,
+                 extract_method='uni_6', seed=0)
+    out = ExtractCLIP(cfg, external_call=True)(torch.arange(1))[0]
+    f = out['CLIP-ViT-B/32']
+    assert list(f.shape) == gold['shape']
+    np.testing.assert_allclose(f[0, :8], gold['first_row_8'],
+                               atol=2e-3, rtol=1e-3)
+    assert abs(float(f.mean()) - gold['mean']) < 1e-3
+    assert abs(float(f.std()) - gold['std']) < 1e-3
+    np.testing.assert_allclose(out['timestamps_ms'], gold['timestamps_ms'])
