@@ -84,6 +84,7 @@ def test_clip_golden_regression(tmp_path):
     import os
     from tests.conftest import synthetic_frames
     from video_features_amd.io.y4m import write_y4m
+    from video_features_amd.extractors.clip import ExtractCLIP
     gold = json.load(open(os.path.join(os.path.dirname(__file__),
                                        'golden_clip.json')))
     vid = str(tmp_path / 'v.y4m')
