@@ -1,0 +1,76 @@
+"""Property-based tests (hypothesis) for the pure-logic layers: sampling,
+sliding windows, TF-SAME padding, temporal merge geometry."""
+import numpy as np
+import torch
+from hypothesis import given, settings, strategies as st
+
+from video_features_amd.io.sampling import (form_slices, sample_indices,
+                                            timestamps_ms)
+from video_features_amd.models.i3d import _same_pad_1d
+
+
+@settings(max_examples=200, deadline=None)
+@given(n=st.integers(1, 5000), k=st.integers(1, 64))
+def test_uni_sampling_invariants(n, k):
+    idxs = sample_indices(f'uni_{k}', n, fps=25.0)
+    assert len(idxs) == k
+    assert all(0 <= i < n for i in idxs)
+    assert list(idxs) == sorted(idxs)
+    ts = timestamps_ms(idxs, 25.0)
+    assert all(t >= 0 for t in ts)
+
+
+@settings(max_examples=200, deadline=None)
+@given(n=st.integers(2, 5000), fps=st.floats(1.0, 120.0),
+       target=st.floats(0.5, 60.0))
+def test_fix_sampling_invariants(n, fps, target):
+    idxs = sample_indices(f'fix_{target}', n, fps=fps)
+    assert len(idxs) >= 1
+    assert all(0 <= i < n for i in idxs)
+    assert list(idxs) == sorted(idxs)
+
+
+@settings(max_examples=200, deadline=None)
+@given(size=st.integers(0, 2000), stack=st.integers(1, 128),
+       step=st.integers(1, 128))
+def test_form_slices_invariants(size, stack, step):
+    """(size-stack)//step+1 full windows, each exactly stack long, inside
+    bounds, strided by step (reference utils/utils.py:117-126)."""
+    slices = form_slices(size, stack, step)
+    expect = max(0, (size - stack) // step + 1) if size >= stack else 0
+    assert len(slices) == expect
+    for i, (a, b) in enumerate(slices):
+        assert b - a == stack and 0 <= a and b <= size
+        assert a == i * step
+
+
+@settings(max_examples=300, deadline=None)
+@given(n=st.integers(1, 500), k=st.integers(1, 9), s=st.integers(1, 4))
+def test_tf_same_pad_output_geometry(n, k, s):
+    """TF-SAME: out = ceil(n/s) for any (n, k, s); front pad <= back pad."""
+    p0, p1 = _same_pad_1d(n, k, s)
+    out = (n + p0 + p1 - k) // s + 1
+    assert out == -(-n // s)
+    assert p0 <= p1 <= p0 + 1 or (p0 == 0 and p1 == 0) or p1 >= p0
+
+
+@settings(max_examples=50, deadline=None)
+@given(b=st.integers(1, 3), t=st.integers(1, 12), o=st.integers(1, 8),
+       st_=st.integers(1, 2))
+def test_temporal_merge_matches_conv1d(b, t, o, st_):
+    """temporal_merge == a true temporal conv with 3 taps (zero pad 1)."""
+    from video_features_amd.models._flat3d import temporal_merge
+    torch.manual_seed(0)
+    h = w = 2
+    y = torch.randn(b * t, 3 * o, h, w)
+    merged = temporal_merge(y, b, kt=3, st=st_, p0=1)
+    # reference: explicit conv1d over t with identity per-tap weights
+    y5 = y.view(b, t, 3, o, h, w)
+    tpad = torch.zeros(b, t + 2, 3, o, h, w)
+    tpad[:, 1:t + 1] = y5
+    to = (t + 2 - 3) // st_ + 1
+    ref = torch.zeros(b, to, o, h, w)
+    for j in range(to):
+        for dt in range(3):
+            ref[:, j] += tpad[:, j * st_ + dt, dt]
+    assert torch.allclose(merged.view(b, to, o, h, w), ref, atol=1e-5)

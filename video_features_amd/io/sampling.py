@@ -55,7 +55,10 @@ def sample_indices(extract_method: str, frame_cnt: int, fps: float) -> np.ndarra
     very short videos — same as the reference's linspace behaviour)."""
     n = num_samples(extract_method, frame_cnt, fps)
     hi = max(frame_cnt - 2, 1)
-    return np.linspace(1, hi, n).astype(np.int64)
+    # clamp into range: the reference's uncapped linspace indexes frame 1
+    # even for a 1-frame video (found by the hypothesis property tests)
+    return np.clip(np.linspace(1, hi, n).astype(np.int64), 0,
+                   frame_cnt - 1)
 
 
 def timestamps_ms(indices: np.ndarray, fps: float) -> List[float]:
