@@ -8,15 +8,18 @@ rank per GPU (RCCL); ranks read RANK/LOCAL_RANK/WORLD_SIZE/MASTER_* from the
 environment.  Work is synthetic (no network): random-init ViT-B/32 weights,
 random uint8 224x224 RGB frames, 12 frames per "video" (the ``uni_12``
 sampling config of BASELINE.json).  Each step runs the full per-batch GPU
-pipeline — H2D copy of uint8 frames, CLIP normalization, bf16 ViT forward
-through the hand-written HIP ops (LayerNorm / QuickGELU / fused MHSA), and
-the D2H feature pull that the extractor performs per video.
+pipeline — double-buffered H2D upload of uint8 frames on a copy stream,
+fused u8→CHW normalization, bf16 ViT forward through the hand-written HIP
+ops (fused LayerNorm(+residual), packed-qkv MFMA flash attention, the
+fc1+QuickGELU fused MFMA GEMM), hipGraph replay per chunk, and the D2H
+feature pull that the extractor performs per video batch.
 
 Weak scaling: per-GPU work is fixed (``--videos-per-step`` per rank);
 ``value`` is the WHOLE-JOB frames/sec aggregated over all ranks.
 
-Secondary config (BASELINE.json config 4): ``--model i3d_raft`` measures
-clips/sec of I3D rgb+flow with RAFT flow on 64-frame stacks.
+Other BASELINE configs: ``--model i3d_raft`` (config 4; ``--flow pwc``
+for the PWC variant), ``--model resnet50`` (config 3), ``--model
+vggish_r21d`` (config 5).
 """
 from __future__ import annotations
 
