@@ -155,3 +155,18 @@ def test_clip_vit_gpu_matches_cpu_fp32(dev):
         out = mg.encode_image(x.to(dev, torch.bfloat16)).float().cpu()
     cos = torch.nn.functional.cosine_similarity(out, ref).min().item()
     assert cos > 0.99, cos
+
+
+def test_clip_vit_b16_gpu_matches_cpu_fp32(dev):
+    """ViT-B/16 = 197 tokens: covers the multi-KV-tile flash path in a
+    full-model forward."""
+    from video_features_amd.models.clip_vit import build_clip_vit
+    torch.manual_seed(0)
+    m = build_clip_vit('CLIP-ViT-B/16').eval()
+    x = torch.randn(2, 3, 224, 224)
+    with torch.no_grad():
+        ref = m.encode_image(x)
+        out = m.to(dev).encode_image(x.to(dev)).cpu()
+    cos = torch.nn.functional.cosine_similarity(out.flatten(), ref.flatten(),
+                                                dim=0).item()
+    assert cos > 0.999, cos
