@@ -61,7 +61,8 @@ def cl_empty(bt: int, c: int, h: int, w: int, like: torch.Tensor):
 
 
 def temporal_merge(y: torch.Tensor, b: int, kt: int, st: int = 1,
-                   p0: int = 1, bias_tap: int = None) -> torch.Tensor:
+                   p0: int = 1, bias_tap: int = None,
+                   relu: bool = False) -> torch.Tensor:
     """y (B*T, kt*O, H, W): temporal-tap conv outputs stacked along channels
     → (B*T', O, H, W) with out[to] = Σ_dt y_dt[to*st - p0 + dt] (zero
     temporal padding).  ``bias_tap``: the tap that already carries the conv
@@ -73,7 +74,7 @@ def temporal_merge(y: torch.Tensor, b: int, kt: int, st: int = 1,
     to = (t + 2 * p0 - kt) // st + 1
     if bias_tap is None:
         bias_tap = kt // 2
-    fused = ops.temporal_merge_fused(y, b, kt, st, p0)
+    fused = ops.temporal_merge_fused(y, b, kt, st, p0, relu)
     if fused is not None:
         return fused
     out = cl_empty(b * to, o, h, w, y)
@@ -94,7 +95,7 @@ def temporal_merge(y: torch.Tensor, b: int, kt: int, st: int = 1,
         s0 = j_lo * st - p0 + dt
         o5[:, j_lo:j_hi + 1] += \
             y5[:, s0:s0 + (j_hi - j_lo) * st + 1:st, dt * o:(dt + 1) * o]
-    return out
+    return torch.relu_(out) if relu else out
 
 
 def temporal_select(xf: torch.Tensor, b: int, st: int) -> torch.Tensor:

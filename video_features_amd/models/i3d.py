@@ -123,6 +123,10 @@ class Unit3D(nn.Module):
                 bcat = None
             y = F.conv2d(xf, wcat, bcat,
                          padding=(self.kernel[1] // 2, self.kernel[2] // 2))
+            if self.activation and not isinstance(self.bn, nn.BatchNorm3d):
+                # BN folded: ReLU rides the merge kernel's epilogue
+                return temporal_merge(y, b, kt=3, st=1, p0=1, bias_tap=1,
+                                      relu=True)
             x = temporal_merge3(y, b)
         if isinstance(self.bn, nn.BatchNorm3d):
             # BatchNorm3d == BatchNorm2d per channel on the flattened view

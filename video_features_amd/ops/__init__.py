@@ -378,13 +378,14 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
 
 
 def temporal_merge_fused(y: torch.Tensor, b: int, kt: int, st: int,
-                         p0: int) -> Optional[torch.Tensor]:
+                         p0: int, relu: bool = False) -> Optional[torch.Tensor]:
     """One-kernel temporal tap merge for the flattened-time conv3d
-    decomposition (see models/_flat3d.py).  Returns None when the HIP path
-    does not apply (caller falls back to the strided-add composition)."""
+    decomposition (see models/_flat3d.py), with optional fused ReLU.
+    Returns None when the HIP path does not apply (caller falls back to
+    the strided-add composition)."""
     if (_use_hip(y)
             and y.is_contiguous(memory_format=torch.channels_last)):
-        return _ext.temporal_merge(y, b, kt, st, p0)
+        return _ext.temporal_merge(y, b, kt, st, p0, relu)
     return None
 
 

@@ -47,7 +47,7 @@ void vfa_maxpool2d_same(const void*, void*, long long, int, int, int, int,
 void vfa_linear_act(const void*, const void*, const void*, const void*,
                     void*, int, int, int, int, hipStream_t);
 void vfa_temporal_merge(const void*, void*, int, int, int, int, int, int,
-                        int, long long, int, hipStream_t);
+                        int, long long, int, int, hipStream_t);
 }
 
 namespace {
@@ -411,7 +411,7 @@ torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
 }
 
 torch::Tensor temporal_merge(torch::Tensor y, int64_t b, int64_t kt,
-                             int64_t st, int64_t p0) {
+                             int64_t st, int64_t p0, bool relu) {
   // y (B*T, kt*O, H, W) channels_last -> (B*T', O, H, W) channels_last
   TORCH_CHECK(y.is_cuda() && y.dim() == 4);
   TORCH_CHECK(cl_contig(y), "channels_last expected");
@@ -422,8 +422,8 @@ torch::Tensor temporal_merge(torch::Tensor y, int64_t b, int64_t kt,
   auto out = torch::empty({(long)b * to, h, w, o}, y.options())
                  .permute({0, 3, 1, 2});
   vfa_temporal_merge(y.data_ptr(), out.data_ptr(), (int)b, t, to, (int)kt,
-                     (int)st, (int)p0, o, (long long)h * w, dtype_tag(y),
-                     current_stream());
+                     (int)st, (int)p0, o, (long long)h * w, relu ? 1 : 0,
+                     dtype_tag(y), current_stream());
   return out;
 }
 

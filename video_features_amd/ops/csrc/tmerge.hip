@@ -9,7 +9,7 @@
 
 namespace {
 
-template <typename T>
+template <typename T, bool RELU>
 __global__ void temporal_merge_kernel(const T* __restrict__ y,
                                       T* __restrict__ out, int b, int t,
                                       int to, int kt, int st, int p0, int o,
@@ -31,7 +31,7 @@ __global__ void temporal_merge_kernel(const T* __restrict__ y,
       acc += to_f32<T>(y[(((long long)bi * t + src) * hw + p) * cin +
                          dt * o + c]);
     }
-    out[i] = from_f32<T>(acc);
+    out[i] = from_f32<T>(RELU ? fmaxf(acc, 0.f) : acc);
   }
 }
 
@@ -41,13 +41,18 @@ extern "C" {
 
 void vfa_temporal_merge(const void* y, void* out, int b, int t, int to,
                         int kt, int st, int p0, int o, long long hw,
-                        int dtype, hipStream_t stream) {
+                        int relu, int dtype, hipStream_t stream) {
   const long long total = (long long)b * to * hw * o;
   const int grid = (int)min((total + 255) / 256, (long long)65536);
 #define VFA_TM_CASE(T)                                                        \
-  hipLaunchKernelGGL((temporal_merge_kernel<T>), dim3(grid), dim3(256), 0,    \
-                     stream, (const T*)y, (T*)out, b, t, to, kt, st, p0, o,   \
-                     hw);
+  if (relu)                                                                   \
+    hipLaunchKernelGGL((temporal_merge_kernel<T, true>), dim3(grid),          \
+                       dim3(256), 0, stream, (const T*)y, (T*)out, b, t, to,  \
+                       kt, st, p0, o, hw);                                    \
+  else                                                                        \
+    hipLaunchKernelGGL((temporal_merge_kernel<T, false>), dim3(grid),         \
+                       dim3(256), 0, stream, (const T*)y, (T*)out, b, t, to,  \
+                       kt, st, p0, o, hw);
   switch (dtype) {
     case VFA_F32: VFA_TM_CASE(float) break;
     case VFA_BF16: VFA_TM_CASE(__hip_bfloat16) break;
