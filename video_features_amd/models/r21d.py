@@ -47,10 +47,11 @@ class Conv2Plus1D(nn.Module):
     def forward(self, x):
         return self.temporal(self.relu(self.bn(self.spatial(x))))
 
-    def forward_flat(self, xf, b):
+    def forward_flat(self, xf, b, relu_after: bool = False):
         """(B*T, C, H, W) channels_last path (see models/_flat3d.py):
         spatial (1,3,3) conv = conv2d; temporal (3,1,1) conv = merged
-        3-tap 1x1 conv2d + shifted (strided) temporal add."""
+        3-tap 1x1 conv2d + shifted (strided) temporal add (with the
+        follow-up ReLU fused into the merge when requested)."""
         ss = self.spatial.stride[1]
         sw = cached_cl_weight(self, 'sw', self.spatial.weight,
                               lambda: self.spatial.weight[:, :, 0])
@@ -71,7 +72,8 @@ class Conv2Plus1D(nn.Module):
         else:
             bcat = None
         y = F.conv2d(y, wcat, bcat)
-        return temporal_merge(y, b, kt=3, st=self.temporal.stride[0], p0=1)
+        return temporal_merge(y, b, kt=3, st=self.temporal.stride[0], p0=1,
+                              relu=relu_after)
 
 
 class R21DBlock(nn.Module):
@@ -111,8 +113,10 @@ class R21DBlock(nn.Module):
             identity = F.conv2d(identity, dw, conv.bias,
                                 stride=conv.stride[1])
             identity = _flat_bn_relu(identity, bn, False)
-        out = self.conv1.forward_flat(xf, b)
-        out = _flat_bn_relu(out, self.bn1, True)
+        bn1_folded = isinstance(self.bn1, nn.Identity)
+        out = self.conv1.forward_flat(xf, b, relu_after=bn1_folded)
+        if not bn1_folded:
+            out = _flat_bn_relu(out, self.bn1, True)
         out = self.conv2.forward_flat(out, b)
         out = _flat_bn_relu(out, self.bn2, False)
         return F.relu(out + identity, inplace=True)
@@ -162,8 +166,11 @@ class R2Plus1D18(nn.Module):
             bcat = cached_cl_weight(self, 's1b', sc1.bias, mk_bcat)
         else:
             bcat = None
-        xf = temporal_merge(F.conv2d(xf, wcat, bcat), b, kt=3, st=1, p0=1)
-        xf = _flat_bn_relu(xf, sbn1, True)
+        sbn1_folded = isinstance(sbn1, nn.Identity)
+        xf = temporal_merge(F.conv2d(xf, wcat, bcat), b, kt=3, st=1, p0=1,
+                            relu=sbn1_folded)
+        if not sbn1_folded:
+            xf = _flat_bn_relu(xf, sbn1, True)
         for layer in (self.layer1, self.layer2, self.layer3, self.layer4):
             for blk in layer:
                 xf = blk.forward_flat(xf, b)
