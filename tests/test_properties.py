@@ -74,3 +74,40 @@ def test_temporal_merge_matches_conv1d(b, t, o, st_):
         for dt in range(3):
             ref[:, j] += tpad[:, j * st_ + dt, dt]
     assert torch.allclose(merged.view(b, to, o, h, w), ref, atol=1e-5)
+
+
+@settings(max_examples=30, deadline=None)
+@given(b=st.integers(1, 2), c=st.integers(1, 8), h=st.integers(3, 12),
+       w=st.integers(3, 12))
+def test_pwc_correlation_cpu_matches_naive(b, c, h, w):
+    """The vectorized CPU cost-volume (the GPU kernel's reference) against
+    a literal per-displacement loop."""
+    from video_features_amd.ops import _pwc_correlation_torch
+    torch.manual_seed(0)
+    f1 = torch.randn(b, c, h, w)
+    f2 = torch.randn(b, c, h, w)
+    out = _pwc_correlation_torch(f1, f2, 4)
+    assert out.shape == (b, 81, h, w)
+    f2p = torch.nn.functional.pad(f2, (4, 4, 4, 4))
+    # channel i = dy*9+dx ordering
+    i = 0
+    for dy in range(9):
+        for dx in range(9):
+            ref = (f1 * f2p[:, :, dy:dy + h, dx:dx + w]).mean(1)
+            assert torch.allclose(out[:, i], ref, atol=1e-5)
+            i += 1
+
+
+@settings(max_examples=50, deadline=None)
+@given(n=st.integers(1, 200), bs=st.integers(1, 16), tp=st.integers(1, 4))
+def test_temporal_shard_partition(n, bs, tp):
+    """Round-robin window shards across tp ranks partition the start list
+    exactly (no loss, no overlap) — the temporal-parallel invariant."""
+    starts = list(range(0, n * 10, 10))[:n]
+    shards = [starts[r::tp] for r in range(tp)]
+    merged = [None] * n
+    for r, sh in enumerate(shards):
+        for j, s in enumerate(sh):
+            assert merged[j * tp + r] is None
+            merged[j * tp + r] = s
+    assert merged == starts
