@@ -99,3 +99,47 @@ def test_clip_golden_regression(tmp_path):
     assert abs(float(f.mean()) - gold['mean']) < 1e-3
     assert abs(float(f.std()) - gold['std']) < 1e-3
     np.testing.assert_allclose(out['timestamps_ms'], gold['timestamps_ms'])
+
+
+def test_decode_pipeline_matches_serial(tmp_path):
+    """The one-video-ahead decode thread must not change results (multi-
+    video run vs per-video serial runs)."""
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    paths = []
+    for i in range(3):
+        p = str(tmp_path / f'v{i}.y4m')
+        write_y4m(p, synthetic_frames(t=10 + i, h=64, w=64, seed=i), fps=25.0)
+        paths.append(p)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=paths, cpu=True,
+                 extract_method='uni_3', seed=0)
+    ex = ExtractCLIP(cfg, external_call=True)
+    piped = ex(torch.arange(3))
+    for i, p in enumerate(paths):
+        solo = ExtractCLIP(cfg.replace(video_paths=[p]),
+                           external_call=True)(torch.arange(1))[0]
+        np.testing.assert_allclose(piped[i]['CLIP-ViT-B/32'],
+                                   solo['CLIP-ViT-B/32'], rtol=1e-5,
+                                   atol=1e-6)
+
+
+def test_decode_pipeline_error_isolation(tmp_path):
+    """A video whose prefetched decode fails must be skipped without
+    poisoning the next video's result."""
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    good1 = str(tmp_path / 'a.y4m')
+    bad = str(tmp_path / 'b.y4m')
+    good2 = str(tmp_path / 'c.y4m')
+    write_y4m(good1, synthetic_frames(t=8, h=64, w=64), fps=25.0)
+    open(bad, 'wb').write(b'NOT A VIDEO AT ALL' * 10)
+    write_y4m(good2, synthetic_frames(t=8, h=64, w=64, seed=2), fps=25.0)
+    cfg = Config(feature_type='CLIP-ViT-B/32',
+                 video_paths=[good1, bad, good2], cpu=True,
+                 extract_method='uni_2')
+    out = ExtractCLIP(cfg, external_call=True)(torch.arange(3))
+    assert len(out) == 2   # bad one skipped, both good ones extracted
+    for fd in out:
+        assert fd['CLIP-ViT-B/32'].shape == (2, 512)

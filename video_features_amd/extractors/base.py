@@ -61,8 +61,14 @@ class BaseExtractor(torch.nn.Module):
         raise NotImplementedError
 
     def extract(self, device: torch.device, models: Any,
-                video_path) -> Dict[str, np.ndarray]:
+                video_path, prepared: Any = None) -> Dict[str, np.ndarray]:
         raise NotImplementedError
+
+    # optional hook: CPU-side decode/sampling for ``video_path``, run one
+    # video AHEAD on a worker thread so host decode overlaps GPU compute
+    # (the decode pipeline of SURVEY §7 hard-part (d)).  Extractors that
+    # override it receive the result as ``prepared``.
+    prepare = None
 
     @staticmethod
     def load_weights(model: torch.nn.Module, path: str) -> None:
@@ -126,14 +132,32 @@ class BaseExtractor(torch.nn.Module):
                                   device)
         models = self.models_for(device)
         feats_list: List[Dict[str, np.ndarray]] = []
-        for idx in indices.tolist():
+        idx_list = indices.tolist()
+        pipeline = type(self).prepare is not None and len(idx_list) > 1
+        pool = fut = None
+        if pipeline:
+            from concurrent.futures import ThreadPoolExecutor
+            pool = ThreadPoolExecutor(max_workers=1,
+                                      thread_name_prefix='vfa-decode')
+        for pos, idx in enumerate(idx_list):
             video_path = self.path_list[idx]
             try:
                 if (self.cfg.resume and not self.external_call
                         and self._already_done(video_path)):
                     self.progress.update()
                     continue
-                feats_dict = self.extract(device, models, video_path)
+                prepared = None
+                if pipeline:
+                    pending, fut = fut, None   # consume BEFORE .result():
+                    # a raise must not replay on the next video
+                    with self._prof('decode'):
+                        prepared = (pending.result() if pending is not None
+                                    else self.prepare(video_path))
+                    if pos + 1 < len(idx_list):
+                        nxt = self.path_list[idx_list[pos + 1]]
+                        fut = pool.submit(self.prepare, nxt)
+                feats_dict = self.extract(device, models, video_path,
+                                          prepared)
                 if self.external_call:
                     feats_list.append(feats_dict)
                 else:
@@ -150,6 +174,8 @@ class BaseExtractor(torch.nn.Module):
                       'Continuing extraction')
                 traceback.print_exc()
             self.progress.update()
+        if pool is not None:
+            pool.shutdown(wait=False, cancel_futures=True)
         rep = self.prof.report(f'{self.feature_type} shard on {device}')
         if rep:
             print(rep)

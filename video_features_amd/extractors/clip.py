@@ -33,13 +33,19 @@ class ExtractCLIP(BaseExtractor):
         model = model.to(device=device, dtype=dtype).eval()
         return model
 
+    def prepare(self, video_path):
+        """CPU decode + frame sampling — runs one video ahead on the
+        decode thread (overlaps the previous video's GPU forward)."""
+        reader = open_video(video_path, self.tmp_path, self.extraction_fps)
+        fps = reader.fps
+        idxs = sample_indices(self.extract_method, reader.frame_count, fps)
+        return torch.from_numpy(reader.read_frames(idxs)), fps, idxs
+
     def extract(self, device: torch.device, model,
-                video_path) -> Dict[str, np.ndarray]:
+                video_path, prepared=None) -> Dict[str, np.ndarray]:
         with self._prof('decode'):
-            reader = open_video(video_path, self.tmp_path, self.extraction_fps)
-            fps = reader.fps
-            idxs = sample_indices(self.extract_method, reader.frame_count, fps)
-            frames_u8 = torch.from_numpy(reader.read_frames(idxs))
+            frames_u8, fps, idxs = (prepared if prepared is not None
+                                    else self.prepare(video_path))
         with self._prof('preprocess'):
             batch = T.clip_preprocess(frames_u8, model.cfg.input_resolution)
         dtype = self.compute_dtype(device)

@@ -122,13 +122,18 @@ class ExtractI3D(BaseExtractor):
         x = T.resize_improved(flow, RESIZE_SIDE, smaller_edge=True)
         return x.to(device=device, dtype=dtype)
 
+    def prepare(self, video_path):
+        vid = video_path[0] if isinstance(video_path, tuple) else video_path
+        return self._read_resized_frames(vid)
+
     # ------------------------------------------------------------ extract
     def extract(self, device: torch.device, models,
-                video_path) -> Dict[str, np.ndarray]:
+                video_path, prepared=None) -> Dict[str, np.ndarray]:
         precomputed = isinstance(video_path, tuple)
         vid_path = video_path[0] if precomputed else video_path
         with self._prof('decode'):
-            frames, fps = self._read_resized_frames(vid_path)
+            frames, fps = (prepared if prepared is not None
+                           else self._read_resized_frames(vid_path))
         dtype = self.compute_dtype(device)
         n = frames.shape[0]
         feats: Dict[str, List] = {s: [] for s in self.streams}

@@ -37,7 +37,7 @@ class ExtractPWC(BaseExtractor):
         return x
 
     def extract(self, device: torch.device, model,
-                video_path) -> Dict[str, np.ndarray]:
+                video_path, prepared=None) -> Dict[str, np.ndarray]:
         reader = open_video(video_path, self.tmp_path, self.extraction_fps)
         fps = reader.fps
         n = reader.frame_count

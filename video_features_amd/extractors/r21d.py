@@ -34,11 +34,18 @@ class ExtractR21D(BaseExtractor):
             self.load_weights(model, self.cfg.weights_path)
         return model.to(device=device, dtype=dtype).eval()
 
-    def extract(self, device: torch.device, model,
-                video_path) -> Dict[str, np.ndarray]:
+    def prepare(self, video_path):
+        # whole-video decode (the reference reads the full video too,
+        # reference extract_r21d.py:102) on the decode thread
         reader = open_video(video_path, self.tmp_path, None)
-        fps = reader.fps
         n = reader.frame_count
+        frames = torch.from_numpy(reader.read_frames(range(n)))
+        return frames, reader.fps, n
+
+    def extract(self, device: torch.device, model,
+                video_path, prepared=None) -> Dict[str, np.ndarray]:
+        all_frames, fps, n = (prepared if prepared is not None
+                              else self.prepare(video_path))
         slices = form_slices(n, self.stack_size, self.step_size)
         if not slices:
             slices = [(0, n)]   # shorter than one stack: use what exists
@@ -47,7 +54,7 @@ class ExtractR21D(BaseExtractor):
         dtype = self.compute_dtype(device)
         feats, ts = [], []
         for (start, end) in slices:
-            frames_u8 = torch.from_numpy(reader.read_frames(range(start, end)))
+            frames_u8 = all_frames[start:end]
             clip = T.r21d_preprocess(frames_u8)[None]        # (1, 3, T, 112, 112)
             clip = clip.to(device=device, dtype=dtype, non_blocking=True)
             feats.append(model.forward_features(clip).float().cpu())

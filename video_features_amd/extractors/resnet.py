@@ -31,7 +31,7 @@ class ExtractResNet(BaseExtractor):
         return model.to(device=device, dtype=dtype).eval()
 
     def extract(self, device: torch.device, model,
-                video_path) -> Dict[str, np.ndarray]:
+                video_path, prepared=None) -> Dict[str, np.ndarray]:
         reader = open_video(video_path, self.tmp_path, self.extraction_fps)
         fps = reader.fps
         dtype = self.compute_dtype(device)

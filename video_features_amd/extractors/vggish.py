@@ -27,14 +27,21 @@ class ExtractVGGish(BaseExtractor):
             self.load_weights(model, self.cfg.weights_path)
         return model.to(device=device, dtype=dtype).eval()
 
+    def prepare(self, video_path):
+        # audio demux + resample + log-mel framing: the CPU-heavy half of
+        # the VGGish pipeline, run one video ahead on the decode thread
+        samples, sr, tmp_files = load_audio_for_video(
+            video_path, self.tmp_path, self.keep_tmp_files)
+        samples = resample(samples, sr, SAMPLE_RATE)
+        wav = torch.from_numpy(np.ascontiguousarray(samples))
+        return waveform_to_examples(wav, None), tmp_files
+
     def extract(self, device: torch.device, model,
-                video_path) -> Dict[str, np.ndarray]:
-        samples, sr, tmp_files = load_audio_for_video(video_path, self.tmp_path,
-                                                      self.keep_tmp_files)
+                video_path, prepared=None) -> Dict[str, np.ndarray]:
+        examples, tmp_files = (prepared if prepared is not None
+                               else self.prepare(video_path))
         try:
-            samples = resample(samples, sr, SAMPLE_RATE)
-            wav = torch.from_numpy(np.ascontiguousarray(samples))
-            examples = waveform_to_examples(wav, device)
+            examples = examples.to(device)
             dtype = self.compute_dtype(device)
             feats = model(examples.to(dtype)).float().cpu().numpy()
         finally:
