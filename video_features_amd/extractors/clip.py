@@ -47,6 +47,11 @@ class ExtractCLIP(BaseExtractor):
             frames_u8, fps, idxs = (prepared if prepared is not None
                                     else self.prepare(video_path))
         with self._prof('preprocess'):
+            # the transforms are pure torch — run them ON the GPU (the u8
+            # upload is 4-12x smaller than uploading preprocessed floats,
+            # and CPU-side bicubic was 87% of end-to-end extractor wall)
+            if device.type == 'cuda':
+                frames_u8 = frames_u8.to(device, non_blocking=True)
             batch = T.clip_preprocess(frames_u8, model.cfg.input_resolution)
         dtype = self.compute_dtype(device)
         feats = []

@@ -55,8 +55,9 @@ class ExtractR21D(BaseExtractor):
         feats, ts = [], []
         for (start, end) in slices:
             frames_u8 = all_frames[start:end]
-            clip = T.r21d_preprocess(frames_u8)[None]        # (1, 3, T, 112, 112)
-            clip = clip.to(device=device, dtype=dtype, non_blocking=True)
+            if device.type == 'cuda':   # preprocess on the GPU (u8 upload)
+                frames_u8 = frames_u8.to(device, non_blocking=True)
+            clip = T.r21d_preprocess(frames_u8)[None].to(dtype)
             feats.append(model.forward_features(clip).float().cpu())
             ts.append(start / fps * 1000.0)
             if self.show_pred:

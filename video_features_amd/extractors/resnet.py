@@ -40,8 +40,9 @@ class ExtractResNet(BaseExtractor):
         for start in range(0, n, self.batch_size):
             idxs = list(range(start, min(start + self.batch_size, n)))
             frames_u8 = torch.from_numpy(reader.read_frames(idxs))
-            batch = T.imagenet_preprocess(frames_u8)
-            batch = batch.to(device=device, dtype=dtype, non_blocking=True)
+            if device.type == 'cuda':   # preprocess on the GPU (u8 upload)
+                frames_u8 = frames_u8.to(device, non_blocking=True)
+            batch = T.imagenet_preprocess(frames_u8).to(dtype)
             feats.append(model.forward_features(batch).float().cpu())
             idx_done.extend(idxs)
             if self.show_pred:
