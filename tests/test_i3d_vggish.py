@@ -205,3 +205,31 @@ def test_i3d_weights_dir_loading(tmp_path):
     assert torch.allclose(
         models['flow_xtr'].update_block.flow_head.conv2.weight,
         raft_ref.update_block.flow_head.conv2.weight)
+
+
+def test_i3d_precomputed_flow_dir(tmp_path):
+    """--flow_type flow with a (video, flow_dir) pair: flow read from
+    flow_x_*.jpg / flow_y_*.jpg frames (reference extract_i3d.py:231-237,
+    266-278 semantics; values map [0,255] -> [-20,20])."""
+    from PIL import Image
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.i3d import ExtractI3D
+    from video_features_amd.io.y4m import write_y4m
+
+    frames = synthetic_frames(t=12, h=64, w=64)
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, frames, fps=25.0)
+    fdir = tmp_path / 'flow'
+    fdir.mkdir()
+    rng = np.random.default_rng(0)
+    for i in range(11):
+        for axis in ('x', 'y'):
+            img = rng.integers(0, 256, (64, 64), dtype=np.uint8)
+            Image.fromarray(img).save(str(fdir / f'flow_{axis}_{i:05d}.jpg'))
+    cfg = Config(feature_type='i3d', video_paths=[vid],
+                 flow_paths=[str(fdir)], cpu=True, stack_size=10,
+                 step_size=10, flow_type='flow', streams=['flow'])
+    ex = ExtractI3D(cfg, external_call=True)
+    out = ex(torch.arange(1))[0]
+    assert out['flow'].shape == (1, 1024)
+    assert np.isfinite(out['flow']).all()
