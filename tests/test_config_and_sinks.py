@@ -126,3 +126,31 @@ def test_main_cli_end_to_end(tmp_path):
     feats = [a for f, a in arrs.items() if 'resnet18' in f
              and 'fps' not in f and 'timestamps' not in f]
     assert feats and feats[0].shape == (20, 512)
+
+
+def test_resume_skips_existing_outputs(tmp_path):
+    """--resume: videos whose outputs already exist are skipped (job-level
+    resume the reference lacks)."""
+    import os
+    import time
+    import torch
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=10, h=64, w=64), fps=25.0)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[vid], cpu=True,
+                 extract_method='uni_2', on_extraction='save_numpy',
+                 output_path=str(tmp_path / 'out'),
+                 tmp_path=str(tmp_path / 'tmp'), resume=True)
+    ExtractCLIP(cfg)(torch.arange(1))
+    out_dir = tmp_path / 'out' / 'CLIP-ViT-B/32'
+    files = sorted(os.listdir(out_dir))
+    assert files
+    mtimes = {f: os.path.getmtime(out_dir / f) for f in files}
+    time.sleep(0.05)
+    ExtractCLIP(cfg)(torch.arange(1))        # resume: must skip
+    for f in files:
+        assert os.path.getmtime(out_dir / f) == mtimes[f], f
+    ExtractCLIP(cfg.replace(resume=False))(torch.arange(1))   # overwrites
+    assert any(os.path.getmtime(out_dir / f) != mtimes[f] for f in files)
