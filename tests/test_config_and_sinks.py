@@ -154,3 +154,18 @@ def test_resume_skips_existing_outputs(tmp_path):
         assert os.path.getmtime(out_dir / f) == mtimes[f], f
     ExtractCLIP(cfg.replace(resume=False))(torch.arange(1))   # overwrites
     assert any(os.path.getmtime(out_dir / f) != mtimes[f] for f in files)
+
+
+def test_save_jpg_flow_sink(tmp_path):
+    """on_extraction=save_jpg writes per-frame Middlebury flow images
+    (works here; the reference's branch is dead, utils/utils.py:96-112)."""
+    import os
+    import numpy as np
+    from video_features_amd.runtime.sinks import action_on_extraction
+    flow = np.random.default_rng(0).normal(0, 3, (4, 2, 16, 20)) \
+        .astype(np.float32)
+    action_on_extraction({'raft': flow, 'fps': np.array(25.0)},
+                         '/x/clip.mp4', str(tmp_path), 'save_jpg',
+                         False, 'raft')
+    files = sorted(os.listdir(tmp_path / 'clip_raft'))
+    assert files == [f'flow_{t:06d}.jpg' for t in range(4)]

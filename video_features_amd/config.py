@@ -23,7 +23,7 @@ FEATURE_TYPES = [
     'vggish_torch',
 ]
 
-ON_EXTRACTION_CHOICES = ['print', 'save_numpy', 'save_pickle']
+ON_EXTRACTION_CHOICES = ['print', 'save_numpy', 'save_pickle', 'save_jpg']
 FLOW_TYPES = ['raft', 'pwc', 'flow']
 
 

@@ -69,6 +69,28 @@ def action_on_extraction(feats_dict: Dict[str, np.ndarray], video_path: str,
                             feature_type, output_direct, 'pkl')
         with open(fpath, 'wb') as f:
             pickle.dump(feats_dict, f)
+    elif on_extraction == 'save_jpg':
+        # flow-image export: (T, 2, H, W) flow fields become per-frame
+        # Middlebury visualizations (the reference documents this sink but
+        # ships a dead branch with an iterate-over-int bug, reference
+        # utils/utils.py:96-112; here it works)
+        from ..utils.flow_viz import flow_to_image
+        from PIL import Image
+        stem = Path(video_path).stem
+        for key, value in feats_dict.items():
+            if key in META_KEYS:
+                continue
+            arr = np.asarray(value)
+            if arr.ndim != 4 or arr.shape[1] != 2:
+                raise ValueError(
+                    f'save_jpg expects (T, 2, H, W) flow features; '
+                    f'{key} has shape {arr.shape} — use save_numpy')
+            vdir = os.path.join(output_path, f'{stem}_{key}')
+            os.makedirs(vdir, exist_ok=True)
+            for t in range(arr.shape[0]):
+                img = flow_to_image(arr[t].transpose(1, 2, 0))
+                Image.fromarray(img).save(
+                    os.path.join(vdir, f'flow_{t:06d}.jpg'))
     else:
         raise ValueError(f'unknown on_extraction {on_extraction!r}')
 
