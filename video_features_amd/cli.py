@@ -29,7 +29,8 @@ def build_parser() -> argparse.ArgumentParser:
     p.add_argument('--tmp_path', default='./tmp')
     p.add_argument('--keep_tmp_files', action='store_true', default=False)
     p.add_argument('--on_extraction', default='print',
-                   choices=['print', 'save_numpy', 'save_pickle'])
+                   choices=['print', 'save_numpy', 'save_pickle',
+                            'save_jpg'])
     p.add_argument('--output_path', default='./output')
     p.add_argument('--output_direct', action='store_true')
     p.add_argument('--extraction_fps', type=float)
