@@ -169,3 +169,40 @@ def test_save_jpg_flow_sink(tmp_path):
                          False, 'raft')
     files = sorted(os.listdir(tmp_path / 'clip_raft'))
     assert files == [f'flow_{t:06d}.jpg' for t in range(4)]
+
+
+def test_tmp_transcode_cleanup(tmp_path):
+    """tmp_path/{stem}.y4m transcode artifacts are removed per video unless
+    --keep_tmp_files."""
+    import os
+    import torch
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'clip1.mp4.y4m')   # source video elsewhere
+    write_y4m(vid, synthetic_frames(t=8, h=64, w=64), fps=25.0)
+    tdir = tmp_path / 'tmp'
+    tdir.mkdir()
+    for keep in (False, True):
+        fake = tdir / 'clip1.mp4.y4m'
+        write_y4m(str(fake), synthetic_frames(t=8, h=64, w=64), fps=25.0)
+        cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[vid],
+                     cpu=True, extract_method='uni_2',
+                     tmp_path=str(tdir), keep_tmp_files=keep)
+        ExtractCLIP(cfg, external_call=True)(torch.arange(1))
+        assert os.path.exists(fake) == keep
+
+
+def test_tmp_cleanup_never_deletes_source(tmp_path):
+    import os
+    import torch
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'v.y4m')           # source INSIDE tmp_path
+    write_y4m(vid, synthetic_frames(t=8, h=64, w=64), fps=25.0)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[vid], cpu=True,
+                 extract_method='uni_2', tmp_path=str(tmp_path))
+    out = ExtractCLIP(cfg, external_call=True)(torch.arange(1))
+    assert len(out) == 1
+    assert os.path.exists(vid)

@@ -173,6 +173,8 @@ class BaseExtractor(torch.nn.Module):
                 print(f'Extraction failed at: {video_path} with error (^). '
                       'Continuing extraction')
                 traceback.print_exc()
+            finally:
+                self._cleanup_tmp(video_path)
             self.progress.update()
         if pool is not None:
             pool.shutdown(wait=False, cancel_futures=True)
@@ -184,6 +186,27 @@ class BaseExtractor(torch.nn.Module):
     def _stem_path(self, video_path) -> str:
         # pre-computed-flow inputs are (video, flow_dir) tuples
         return video_path[0] if isinstance(video_path, tuple) else video_path
+
+    def _cleanup_tmp(self, video_path) -> None:
+        """Remove this video's transcode artifact (tmp_path/{stem}.y4m,
+        written by io.ffmpeg.decode_to_y4m for non-native codecs) unless
+        ``--keep_tmp_files`` (reference extract_resnet.py:159-160
+        semantics; audio tmp wavs are cleaned by the vggish extractor
+        itself)."""
+        if self.keep_tmp_files:
+            return
+        import os
+        from pathlib import Path
+        src = str(self._stem_path(video_path))
+        stem = Path(src).stem
+        p = os.path.join(self.tmp_path, stem + '.y4m')
+        if os.path.abspath(p) == os.path.abspath(src):
+            return          # the SOURCE lives in tmp_path — never delete it
+        if os.path.exists(p):
+            try:
+                os.remove(p)
+            except OSError:
+                pass
 
     def _already_done(self, video_path) -> bool:
         keys = getattr(self, 'output_feat_keys', [self.feature_type])
