@@ -143,3 +143,33 @@ def test_decode_pipeline_error_isolation(tmp_path):
     assert len(out) == 2   # bad one skipped, both good ones extracted
     for fd in out:
         assert fd['CLIP-ViT-B/32'].shape == (2, 512)
+
+
+def test_clip_rn50_extractor(tmp_path):
+    """CLIP-RN50: the ModifiedResNet tower the reference codes but never
+    exposes — here a first-class feature type, (N, 1024) features."""
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=10, h=64, w=64), fps=25.0)
+    cfg = Config(feature_type='CLIP-RN50', video_paths=[vid], cpu=True,
+                 extract_method='uni_2')
+    out = ExtractCLIP(cfg, external_call=True)(torch.arange(1))[0]
+    assert out['CLIP-RN50'].shape == (2, 1024)
+    assert np.isfinite(out['CLIP-RN50']).all()
+
+
+def test_clip_rn50_openai_scheme_roundtrip():
+    """OpenAI RN50 visual-tower key scheme loads via the converter."""
+    from video_features_amd.models.clip_resnet import build_clip_resnet
+    from video_features_amd.utils.convert_checkpoints import convert_auto
+    torch.manual_seed(0)
+    m = build_clip_resnet('CLIP-RN50')
+    sd = m.state_dict()
+    legacy = {'visual.' + k: v.clone() for k, v in sd.items()}
+    legacy['logit_scale'] = torch.zeros(())
+    m2 = build_clip_resnet('CLIP-RN50')
+    m2.load_state_dict(convert_auto(legacy))
+    for k, v in m2.state_dict().items():
+        assert torch.equal(v, sd[k]), k

@@ -20,6 +20,7 @@ FEATURE_TYPES = [
     'resnet18', 'resnet34', 'resnet50', 'resnet101', 'resnet152',
     'raft', 'pwc',
     'CLIP-ViT-B/32', 'CLIP-ViT-B/16', 'CLIP4CLIP-ViT-B-32',
+    'CLIP-RN50', 'CLIP-RN101',
     'vggish_torch',
 ]
 

@@ -16,6 +16,7 @@ import torch
 from .. import transforms as T
 from ..io.sampling import sample_indices, timestamps_ms
 from ..io.video import open_video
+from ..models.clip_resnet import build_clip_resnet
 from ..models.clip_vit import build_clip_vit
 from .base import BaseExtractor
 
@@ -27,7 +28,10 @@ class ExtractCLIP(BaseExtractor):
         self.batch_size = max(1, self.cfg.batch_size)
 
     def build_models(self, device: torch.device, dtype: torch.dtype):
-        model = build_clip_vit(self.feature_type)
+        if self.feature_type.startswith('CLIP-RN'):
+            model = build_clip_resnet(self.feature_type)
+        else:
+            model = build_clip_vit(self.feature_type)
         if self.cfg.weights_path:
             self.load_weights(model, self.cfg.weights_path)
         model = model.to(device=device, dtype=dtype).eval()
@@ -52,7 +56,9 @@ class ExtractCLIP(BaseExtractor):
             # and CPU-side bicubic was 87% of end-to-end extractor wall)
             if device.type == 'cuda':
                 frames_u8 = frames_u8.to(device, non_blocking=True)
-            batch = T.clip_preprocess(frames_u8, model.cfg.input_resolution)
+            res = getattr(model, 'input_resolution', None) \
+                or model.cfg.input_resolution
+            batch = T.clip_preprocess(frames_u8, res)
         dtype = self.compute_dtype(device)
         feats = []
         with self._prof('infer'):
