@@ -98,3 +98,11 @@ def test_vggish_e2e(dev, tmp_path):
                  tmp_path=str(tmp_path / 'tmp'))
     out = _run(cfg, dev)
     assert out['vggish_torch'].shape == (2, 128)
+
+
+def test_clip_rn50_e2e(dev, tmp_path):
+    cfg = Config(feature_type='CLIP-RN50', video_paths=[_vid(tmp_path)],
+                 extract_method='uni_4', tmp_path=str(tmp_path / 'tmp'))
+    out = _run(cfg, dev)
+    f = out['CLIP-RN50']
+    assert f.shape == (4, 1024) and np.isfinite(f).all()
