@@ -206,3 +206,10 @@ def test_tmp_cleanup_never_deletes_source(tmp_path):
     out = ExtractCLIP(cfg, external_call=True)(torch.arange(1))
     assert len(out) == 1
     assert os.path.exists(vid)
+
+
+def test_every_feature_type_has_an_extractor():
+    from video_features_amd.config import FEATURE_TYPES
+    from video_features_amd.models.registry import get_extractor_class
+    for ft in FEATURE_TYPES:
+        assert get_extractor_class(ft) is not None, ft

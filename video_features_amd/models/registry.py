@@ -43,6 +43,8 @@ _REGISTRY: Dict[str, Callable] = {
     'CLIP-ViT-B/32': _clip,
     'CLIP-ViT-B/16': _clip,
     'CLIP4CLIP-ViT-B-32': _clip,
+    'CLIP-RN50': _clip,
+    'CLIP-RN101': _clip,
     'resnet18': _resnet,
     'resnet34': _resnet,
     'resnet50': _resnet,
