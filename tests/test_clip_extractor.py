@@ -173,3 +173,35 @@ def test_clip_rn50_openai_scheme_roundtrip():
     m2.load_state_dict(convert_auto(legacy))
     for k, v in m2.state_dict().items():
         assert torch.equal(v, sd[k]), k
+
+
+def test_resume_skip_does_not_mix_prefetched_decodes(tmp_path):
+    """Regression: with --resume skipping a middle video, the prefetched
+    decode of the skipped video must NOT be consumed by the next one."""
+    import os
+    paths = []
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.io.y4m import write_y4m
+    for i in range(3):
+        p = str(tmp_path / f'v{i}.y4m')
+        write_y4m(p, synthetic_frames(t=10, h=64, w=64, seed=10 + i),
+                  fps=25.0)
+        paths.append(p)
+    out_dir = tmp_path / 'out'
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=paths, cpu=True,
+                 extract_method='uni_2', seed=0, resume=True,
+                 on_extraction='save_numpy', output_path=str(out_dir),
+                 tmp_path=str(tmp_path / 'tmp'))
+    # pre-extract ONLY the middle video so resume skips it
+    ExtractCLIP(cfg.replace(video_paths=[paths[1]]))(torch.arange(1))
+    ExtractCLIP(cfg)(torch.arange(3))
+    # ground truth per video, solo runs
+    for i, p in enumerate(paths):
+        solo = ExtractCLIP(cfg.replace(video_paths=[p],
+                                       on_extraction='print'),
+                           external_call=True)(torch.arange(1))[0]
+        feat_dir = out_dir / 'CLIP-ViT-B/32'
+        got = np.load(feat_dir / f'v{i}_CLIP-ViT-B_32.npy')
+        np.testing.assert_allclose(got, solo['CLIP-ViT-B/32'],
+                                   rtol=1e-5, atol=1e-6, err_msg=p)

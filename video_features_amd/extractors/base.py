@@ -141,15 +141,18 @@ class BaseExtractor(torch.nn.Module):
                                       thread_name_prefix='vfa-decode')
         for pos, idx in enumerate(idx_list):
             video_path = self.path_list[idx]
+            # a pending future (submitted last iteration) is ALWAYS for
+            # THIS video: take ownership first, so a resume-skip or a
+            # raise can never hand video i's decode to video i+1
+            pending, fut = fut, None
             try:
                 if (self.cfg.resume and not self.external_call
                         and self._already_done(video_path)):
+                    pending = None             # discard this video's decode
                     self.progress.update()
                     continue
                 prepared = None
                 if pipeline:
-                    pending, fut = fut, None   # consume BEFORE .result():
-                    # a raise must not replay on the next video
                     with self._prof('decode'):
                         prepared = (pending.result() if pending is not None
                                     else self.prepare(video_path))
