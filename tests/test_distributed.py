@@ -101,3 +101,19 @@ def test_temporal_parallel_rejected_for_framewise():
     with pytest.raises(ValueError):
         sanity_check(Config(feature_type='resnet50',
                             temporal_parallel=True))
+
+
+def test_merge_temporal_shards_ragged():
+    """Rank 0 may hold one more window than the others; the merge must
+    interleave rows back exactly."""
+    from video_features_amd.runtime.dist import merge_temporal_shards
+    a = {'rgb': np.arange(6).reshape(3, 2), 'fps': np.array(25.0),
+         'timestamps_ms': np.array([0.0, 20.0, 40.0])}
+    b = {'rgb': np.arange(100, 104).reshape(2, 2), 'fps': np.array(25.0),
+         'timestamps_ms': np.array([10.0, 30.0])}
+    m = merge_temporal_shards([a, b])
+    np.testing.assert_array_equal(
+        m['rgb'], [[0, 1], [100, 101], [2, 3], [102, 103], [4, 5]])
+    np.testing.assert_array_equal(m['timestamps_ms'],
+                                  [0.0, 10.0, 20.0, 30.0, 40.0])
+    assert float(m['fps']) == 25.0

@@ -35,3 +35,14 @@ def test_serve_extract_roundtrip(tmp_path):
     # second request reuses the resident models
     r2 = client.post('/extract?filename=v.y4m', content=vid.read_bytes())
     assert r2.status_code == 200
+
+
+def test_serve_error_paths(tmp_path):
+    cfg = Config(feature_type='CLIP-ViT-B/32', cpu=True,
+                 extract_method='uni_2', video_paths=['__server__'],
+                 tmp_path=str(tmp_path / 'tmp'))
+    client = TestClient(create_app(cfg))
+    r = client.post('/extract?filename=v.y4m', content=b'')
+    assert r.status_code == 400
+    r = client.post('/extract?filename=v.y4m', content=b'garbage' * 100)
+    assert r.status_code == 422
