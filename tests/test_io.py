@@ -108,3 +108,18 @@ def test_listing(tmp_path, frames16):
     cfg = Config(video_paths=[str(v1), str(tmp_path / 'nope.y4m')])
     with pytest.raises(FileNotFoundError):
         form_list_from_user_input(cfg)
+
+
+def test_y4m_c444_roundtrip(tmp_path):
+    """C444 y4m keeps full-resolution chroma: random-noise frames roundtrip
+    within YUV<->RGB rounding (C420 subsampling would lose ~±240 on sharp
+    chroma edges)."""
+    from video_features_amd.io.y4m import write_y4m
+    rng = np.random.default_rng(0)
+    frames = rng.integers(0, 256, (3, 33, 47, 3), dtype=np.uint8)
+    p = str(tmp_path / 'c444.y4m')
+    write_y4m(p, frames, fps=30.0, colorspace='C444')
+    r = open_video(p)
+    back = r.read_frames([0, 1, 2])
+    err = np.abs(back.astype(int) - frames.astype(int)).max()
+    assert err <= 3, err

@@ -16,11 +16,10 @@ import numpy as np
 _MAGIC = b'YUV4MPEG2'
 
 
-def _yuv420_to_rgb(y: np.ndarray, u: np.ndarray, v: np.ndarray) -> np.ndarray:
-    """BT.601 full-swing-ish conversion used by the JPEG/JFIF convention."""
-    h, w = y.shape
-    u_full = np.repeat(np.repeat(u, 2, axis=0), 2, axis=1)[:h, :w]
-    v_full = np.repeat(np.repeat(v, 2, axis=0), 2, axis=1)[:h, :w]
+def _yuv_to_rgb(y: np.ndarray, u_full: np.ndarray,
+                v_full: np.ndarray) -> np.ndarray:
+    """BT.601 full-swing conversion (JPEG/JFIF convention), full-res
+    chroma planes."""
     yf = y.astype(np.float32)
     uf = u_full.astype(np.float32) - 128.0
     vf = v_full.astype(np.float32) - 128.0
@@ -28,6 +27,24 @@ def _yuv420_to_rgb(y: np.ndarray, u: np.ndarray, v: np.ndarray) -> np.ndarray:
     g = yf - 0.344136 * uf - 0.714136 * vf
     b = yf + 1.772 * uf
     return np.clip(np.stack([r, g, b], axis=-1), 0, 255).astype(np.uint8)
+
+
+def _yuv420_to_rgb(y: np.ndarray, u: np.ndarray, v: np.ndarray) -> np.ndarray:
+    h, w = y.shape
+    u_full = np.repeat(np.repeat(u, 2, axis=0), 2, axis=1)[:h, :w]
+    v_full = np.repeat(np.repeat(v, 2, axis=0), 2, axis=1)[:h, :w]
+    return _yuv_to_rgb(y, u_full, v_full)
+
+
+def _rgb_to_yuv(rgb: np.ndarray):
+    f = rgb.astype(np.float32)
+    r, g, b = f[..., 0], f[..., 1], f[..., 2]
+    y = 0.299 * r + 0.587 * g + 0.114 * b
+    u = -0.168736 * r - 0.331264 * g + 0.5 * b + 128.0
+    v = 0.5 * r - 0.418688 * g - 0.081312 * b + 128.0
+    return (np.clip(y, 0, 255).astype(np.uint8),
+            np.clip(u, 0, 255).astype(np.uint8),
+            np.clip(v, 0, 255).astype(np.uint8))
 
 
 def _rgb_to_yuv420(rgb: np.ndarray) -> Tuple[np.ndarray, np.ndarray, np.ndarray]:
@@ -75,12 +92,18 @@ class Y4MReader:
                 fps_num, fps_den = int(num), int(den)
             elif t.startswith('C'):
                 colorspace = t
-        if not colorspace.startswith('C420'):
-            raise ValueError(f'{path}: only 4:2:0 y4m supported, got {colorspace}')
+        if colorspace.startswith('C420'):
+            self._c444 = False
+        elif colorspace.startswith('C444'):
+            self._c444 = True
+        else:
+            raise ValueError(
+                f'{path}: only 4:2:0 / 4:4:4 y4m supported, got {colorspace}')
         self.fps = fps_num / fps_den
         self._header_len = len(header)
         y_sz = self.width * self.height
-        c_sz = ((self.width + 1) // 2) * ((self.height + 1) // 2)
+        c_sz = y_sz if self._c444 else \
+            ((self.width + 1) // 2) * ((self.height + 1) // 2)
         self._frame_data = y_sz + 2 * c_sz
         self._y_sz, self._c_sz = y_sz, c_sz
         # frame record = b'FRAME...\n' + planes; assume constant FRAME header len
@@ -103,29 +126,38 @@ class Y4MReader:
             f.seek(off)
             raw = f.read(self._frame_data)
         w, h = self.width, self.height
-        cw, ch = (w + 1) // 2, (h + 1) // 2
+        cw, ch = (w, h) if self._c444 else ((w + 1) // 2, (h + 1) // 2)
         y = np.frombuffer(raw, np.uint8, self._y_sz).reshape(h, w)
         u = np.frombuffer(raw, np.uint8, self._c_sz, offset=self._y_sz).reshape(ch, cw)
         v = np.frombuffer(raw, np.uint8, self._c_sz, offset=self._y_sz + self._c_sz).reshape(ch, cw)
+        if self._c444:
+            return _yuv_to_rgb(y, u, v)
         return _yuv420_to_rgb(y, u, v)
 
     def read_frames(self, indices) -> np.ndarray:
         return np.stack([self.read_frame(int(i)) for i in indices])
 
 
-def write_y4m(path: str, frames: np.ndarray, fps: float = 25.0) -> None:
-    """Write (T, H, W, 3) uint8 RGB frames as a 4:2:0 y4m file."""
+def write_y4m(path: str, frames: np.ndarray, fps: float = 25.0,
+              colorspace: str = 'C420jpeg') -> None:
+    """Write (T, H, W, 3) uint8 RGB frames as y4m.  ``C420jpeg`` matches
+    typical video chroma (lossy roundtrip on sharp chroma edges);
+    ``C444`` keeps full-resolution chroma."""
     frames = np.asarray(frames)
     if frames.ndim != 4 or frames.shape[-1] != 3:
         raise ValueError(f'expected (T,H,W,3), got {frames.shape}')
+    c444 = colorspace.startswith('C444')
     t, h, w, _ = frames.shape
     from fractions import Fraction
     fr = Fraction(fps).limit_denominator(1001)
     with open(path, 'wb') as f:
-        f.write(f'YUV4MPEG2 W{w} H{h} F{fr.numerator}:{fr.denominator} Ip A1:1 C420jpeg\n'
-                .encode('ascii'))
+        f.write(f'YUV4MPEG2 W{w} H{h} F{fr.numerator}:{fr.denominator} '
+                f'Ip A1:1 {colorspace}\n'.encode('ascii'))
         for i in range(t):
-            y, u, v = _rgb_to_yuv420(frames[i])
+            if c444:
+                y, u, v = _rgb_to_yuv(frames[i])
+            else:
+                y, u, v = _rgb_to_yuv420(frames[i])
             f.write(b'FRAME\n')
             f.write(y.tobytes())
             f.write(u.tobytes())
