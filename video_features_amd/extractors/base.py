@@ -148,7 +148,18 @@ class BaseExtractor(torch.nn.Module):
             try:
                 if (self.cfg.resume and not self.external_call
                         and self._already_done(video_path)):
-                    pending = None             # discard this video's decode
+                    if pending is not None:
+                        # consume + discard this video's prefetched decode;
+                        # let the extractor release any side artifacts
+                        # (e.g. vggish tmp wavs)
+                        try:
+                            stale = pending.result()
+                        except Exception:
+                            stale = None
+                        pending = None
+                        hook = getattr(self, 'discard_prepared', None)
+                        if stale is not None and hook is not None:
+                            hook(stale)
                     self.progress.update()
                     continue
                 prepared = None

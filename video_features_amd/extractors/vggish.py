@@ -36,6 +36,14 @@ class ExtractVGGish(BaseExtractor):
         wav = torch.from_numpy(np.ascontiguousarray(samples))
         return waveform_to_examples(wav, None), tmp_files
 
+    def discard_prepared(self, prepared) -> None:
+        # a prefetched-then-skipped video's tmp wavs must not leak
+        _, tmp_files = prepared
+        if not self.keep_tmp_files:
+            for f in tmp_files:
+                if os.path.exists(f):
+                    os.remove(f)
+
     def extract(self, device: torch.device, model,
                 video_path, prepared=None) -> Dict[str, np.ndarray]:
         examples, tmp_files = (prepared if prepared is not None
