@@ -43,7 +43,10 @@ class ExtractCLIP(BaseExtractor):
         reader = open_video(video_path, self.tmp_path, self.extraction_fps)
         fps = reader.fps
         idxs = sample_indices(self.extract_method, reader.frame_count, fps)
-        return torch.from_numpy(reader.read_frames(idxs)), fps, idxs
+        frames = torch.from_numpy(reader.read_frames(idxs))
+        if torch.cuda.is_available():
+            frames = frames.pin_memory()     # async-capable H2D upload
+        return frames, fps, idxs
 
     def extract(self, device: torch.device, model,
                 video_path, prepared=None) -> Dict[str, np.ndarray]:

@@ -96,7 +96,10 @@ class ExtractI3D(BaseExtractor):
             idxs = np.linspace(0, n - 1, need).round().astype(np.int64)
         else:
             idxs = np.arange(n)
-        return torch.from_numpy(reader.read_frames(idxs)), reader.fps
+        frames = torch.from_numpy(reader.read_frames(idxs))
+        if torch.cuda.is_available():
+            frames = frames.pin_memory()     # async-capable H2D upload
+        return frames, reader.fps
 
     def _compute_flow(self, models, stacks: torch.Tensor) -> torch.Tensor:
         """(B, S+1, 3, H, W) frames → (B*S, 2, H, W) flow via RAFT or PWC
