@@ -213,3 +213,30 @@ def test_every_feature_type_has_an_extractor():
     from video_features_amd.models.registry import get_extractor_class
     for ft in FEATURE_TYPES:
         assert get_extractor_class(ft) is not None, ft
+
+
+def test_cli_args_roundtrip_to_config():
+    """Every reference CLI flag parses into the Config API unchanged."""
+    from video_features_amd.cli import build_parser
+    args = build_parser().parse_args([
+        '--feature_type', 'i3d', '--video_paths', 'a.mp4', 'b.mp4',
+        '--device_ids', '0', '1', '--flow_type', 'raft',
+        '--streams', 'rgb', 'flow', '--stack_size', '24',
+        '--step_size', '12', '--extraction_fps', '12.5',
+        '--on_extraction', 'save_numpy', '--output_path', '/o',
+        '--tmp_path', '/t', '--keep_tmp_files', '--output_direct',
+        '--batch_size', '4', '--side_size', '288',
+        '--resize_to_larger_edge', '--show_pred', '--temporal_parallel',
+        '--gather_features', '--resume', '--profile', '--seed', '7',
+        '--dtype', 'bf16'])
+    cfg = Config.coerce(args)
+    assert cfg.feature_type == 'i3d' and cfg.video_paths == ['a.mp4', 'b.mp4']
+    assert cfg.device_ids == [0, 1] and cfg.flow_type == 'raft'
+    assert cfg.streams == ['rgb', 'flow']
+    assert cfg.stack_size == 24 and cfg.step_size == 12
+    assert cfg.extraction_fps == 12.5 and cfg.on_extraction == 'save_numpy'
+    assert cfg.keep_tmp_files and cfg.output_direct and cfg.show_pred
+    assert cfg.batch_size == 4 and cfg.side_size == 288
+    assert cfg.resize_to_smaller_edge is False
+    assert cfg.temporal_parallel and cfg.gather_features and cfg.resume
+    assert cfg.profile and cfg.seed == 7 and cfg.dtype == 'bf16'
