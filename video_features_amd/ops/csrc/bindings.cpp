@@ -125,6 +125,8 @@ torch::Tensor pwc_correlation(torch::Tensor f1, torch::Tensor f2,
   TORCH_CHECK(f1.is_cuda() && f1.is_contiguous() && f2.is_contiguous());
   TORCH_CHECK(max_disp == 4, "kernel is specialized for max_disp=4");
   TORCH_CHECK(f1.sizes() == f2.sizes());
+  TORCH_CHECK(f1.size(1) <= 256,
+              "corr_wave registers cover C<=256 (PWC max is 196)");
   const int b = (int)f1.size(0), c = (int)f1.size(1);
   const int h = (int)f1.size(2), w = (int)f1.size(3);
   auto opts = f1.options();
