@@ -114,8 +114,10 @@ def layer_norm_residual(x: torch.Tensor, res: torch.Tensor,
 def preprocess_u8_chw(frames_u8: torch.Tensor, mean, std,
                       bf16: bool = True) -> torch.Tensor:
     """(T, H, W, 3) uint8 → (T, 3, H, W) normalized bf16/f32 in one fused
-    kernel on GPU; eager chain on CPU."""
-    if _use_hip(frames_u8):
+    kernel on GPU; eager chain on CPU (and for H*W not divisible by 4 —
+    the kernel vectorizes 4 pixels per thread)."""
+    if _use_hip(frames_u8) and \
+            (frames_u8.shape[1] * frames_u8.shape[2]) % 4 == 0:
         return _ext.u8_chw_norm(frames_u8.contiguous(), list(map(float, mean)),
                                 list(map(float, std)), bf16)
     x = frames_u8.permute(0, 3, 1, 2).float() / 255.0
