@@ -240,3 +240,18 @@ def test_cli_args_roundtrip_to_config():
     assert cfg.resize_to_smaller_edge is False
     assert cfg.temporal_parallel and cfg.gather_features and cfg.resume
     assert cfg.profile and cfg.seed == 7 and cfg.dtype == 'bf16'
+
+
+def test_baseline_config1_sample_video():
+    """BASELINE.json config 1 verbatim: CLIP-ViT-B/32 --cpu uni_12 on the
+    sample clip → (12, 512) features."""
+    import os
+    import torch
+    from video_features_amd.extractors.clip import ExtractCLIP
+    root = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    vid = os.path.join(root, 'sample', 'v_GGSY1Qvo990.y4m')
+    assert os.path.exists(vid)
+    cfg = Config(feature_type='CLIP-ViT-B/32', video_paths=[vid], cpu=True,
+                 extract_method='uni_12')
+    out = ExtractCLIP(cfg, external_call=True)(torch.arange(1))[0]
+    assert out['CLIP-ViT-B/32'].shape == (12, 512)
