@@ -133,6 +133,14 @@ class BaseExtractor(torch.nn.Module):
         models = self.models_for(device)
         feats_list: List[Dict[str, np.ndarray]] = []
         idx_list = indices.tolist()
+        # multi-process sharding: this worker only sees its own indices, so
+        # the bar total is the shard size, not the whole video list
+        if getattr(self.progress, 'total', None) not in (None, len(idx_list)):
+            try:
+                self.progress.total = len(idx_list)
+                self.progress.refresh()
+            except Exception:
+                pass
         pipeline = type(self).prepare is not None and len(idx_list) > 1
         pool = fut = None
         if pipeline:
