@@ -4,6 +4,11 @@ import numpy as np
 import torch
 from hypothesis import given, settings, strategies as st
 
+# derandomized: the CI gate must be deterministic — a freshly-discovered
+# failing example should come from a dev run, not break the build
+settings.register_profile('ci', derandomize=True)
+settings.load_profile('ci')
+
 from video_features_amd.io.sampling import (form_slices, sample_indices,
                                             timestamps_ms)
 from video_features_amd.models.i3d import _same_pad_1d
