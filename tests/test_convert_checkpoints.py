@@ -50,6 +50,61 @@ def test_clip_visual_scheme_roundtrip():
         assert torch.equal(v, sd[k]), k
 
 
+def test_clip_full_model_dict_with_text_resblocks():
+    """A FULL OpenAI CLIP dict carries text-tower 'transformer.resblocks.*'
+    keys at top level; they must not collide with the visual tower's renamed
+    blocks (reference loads full model dicts, extract_clip.py:46)."""
+    from video_features_amd.models.clip_vit import VisionTransformer
+    torch.manual_seed(0)
+    m = VisionTransformer()
+    sd = m.state_dict()
+    legacy = {}
+    for k, v in sd.items():
+        k = 'visual.' + k.replace('blocks.', 'transformer.resblocks.')
+        k = k.replace('.attn.qkv.weight', '.attn.in_proj_weight')
+        k = k.replace('.attn.qkv.bias', '.attn.in_proj_bias')
+        k = k.replace('.attn.proj.', '.attn.out_proj.')
+        legacy[k] = v.clone()
+    # text tower of a full model: resblocks with DIFFERENT (wrong if loaded)
+    # shapes, plus the other text keys
+    for i in range(12):
+        legacy[f'transformer.resblocks.{i}.attn.in_proj_weight'] = \
+            torch.zeros(3 * 512, 512)
+        legacy[f'transformer.resblocks.{i}.mlp.c_fc.weight'] = \
+            torch.zeros(2048, 512)
+    legacy['positional_embedding'] = torch.zeros(77, 512)
+    legacy['token_embedding.weight'] = torch.zeros(49408, 512)
+    legacy['text_projection'] = torch.zeros(512, 512)
+    legacy['ln_final.weight'] = torch.zeros(512)
+    legacy['logit_scale'] = torch.zeros(())
+    back = cc.convert_auto(legacy)
+    m2 = VisionTransformer()
+    m2.load_state_dict(back)  # strict: text keys must all be gone
+    for k, v in m2.state_dict().items():
+        assert torch.equal(v, sd[k]), k
+
+
+def test_clip_bare_visual_tower_dict():
+    """A visual-tower-only dict (no 'visual.' prefix) keeps its
+    positional_embedding — it belongs to the image encoder there."""
+    from video_features_amd.models.clip_vit import VisionTransformer
+    torch.manual_seed(0)
+    m = VisionTransformer()
+    sd = m.state_dict()
+    legacy = {}
+    for k, v in sd.items():
+        k = k.replace('blocks.', 'transformer.resblocks.')
+        k = k.replace('.attn.qkv.weight', '.attn.in_proj_weight')
+        k = k.replace('.attn.qkv.bias', '.attn.in_proj_bias')
+        k = k.replace('.attn.proj.', '.attn.out_proj.')
+        legacy[k] = v.clone()
+    back = cc.convert_clip_visual(legacy)
+    m2 = VisionTransformer()
+    m2.load_state_dict(back)
+    for k, v in m2.state_dict().items():
+        assert torch.equal(v, sd[k]), k
+
+
 def test_r21d_torchvision_scheme_roundtrip():
     from video_features_amd.models.r21d import R2Plus1D18
     torch.manual_seed(0)

@@ -96,6 +96,31 @@ def test_temporal_parallel_matches_single(tmp_path):
                                rtol=1e-4, atol=1e-4)
 
 
+@pytest.mark.timeout(600)
+def test_temporal_parallel_sink_and_resume(tmp_path):
+    """tp mode sinks per video on rank 0 (bounded memory) and honors
+    --resume: a second run with existing outputs must not rewrite them."""
+    os.environ.setdefault('MASTER_ADDR', '127.0.0.1')
+    from tests.conftest import synthetic_frames
+    from video_features_amd.io.y4m import write_y4m
+    vid = str(tmp_path / 'long.y4m')
+    write_y4m(vid, synthetic_frames(t=25, h=64, w=64), fps=25.0)
+    base = Config(feature_type='i3d', video_paths=[vid], cpu=True,
+                  stack_size=10, step_size=10, flow_type='pwc',
+                  on_extraction='save_numpy',
+                  output_path=str(tmp_path / 'out'),
+                  tmp_path=str(tmp_path / 'tmp'),
+                  device_ids=[0, 1], temporal_parallel=True)
+    run_extraction(base)
+    out_dir = tmp_path / 'out' / 'i3d'
+    files = sorted(os.listdir(out_dir))
+    assert any('rgb' in f for f in files) and any('flow' in f for f in files)
+    mtimes = {f: os.path.getmtime(out_dir / f) for f in files}
+    run_extraction(base.replace(resume=True))
+    for f, t in mtimes.items():
+        assert os.path.getmtime(out_dir / f) == t, f'{f} was rewritten'
+
+
 def test_temporal_parallel_rejected_for_framewise():
     from video_features_amd.config import sanity_check
     with pytest.raises(ValueError):

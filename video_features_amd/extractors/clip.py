@@ -64,9 +64,14 @@ class ExtractCLIP(BaseExtractor):
             batch = T.clip_preprocess(frames_u8, res)
         dtype = self.compute_dtype(device)
         feats = []
+        # reference semantics (extract_clip.py:125-128): the whole sampled
+        # stack goes through encode_image in ONE forward; batch_size is an
+        # additive knob here — the default (1) keeps whole-stack behavior,
+        # an explicit --batch_size B is honored exactly (e.g. to fit memory)
+        bs = self.batch_size if self.batch_size > 1 else batch.shape[0]
         with self._prof('infer'):
-            for s in range(0, batch.shape[0], max(self.batch_size, 16)):
-                chunk = batch[s:s + max(self.batch_size, 16)]
+            for s in range(0, batch.shape[0], bs):
+                chunk = batch[s:s + bs]
                 chunk = chunk.to(device=device, dtype=dtype, non_blocking=True)
                 feats.append(model.encode_image(chunk).float().cpu())
             features = torch.cat(feats).numpy()

@@ -51,10 +51,17 @@ class ExtractI3D(BaseExtractor):
         if not wp:
             return
         p = Path(wp)
-        if p.is_dir():
-            p = p / f'{name}.pt'
-            if not p.exists():
-                p = p.with_suffix('.pth')
+        if not p.is_dir():
+            # a plain file cannot name which of the (up to 4) stream models
+            # it belongs to — loading it into all of them would crash on a
+            # strict load into the wrong net mid-build
+            raise ValueError(
+                f'--weights_path for i3d must be a DIRECTORY containing '
+                f'i3d_rgb.pt / i3d_flow.pt / raft.pth / pwc.pth, got file '
+                f'{wp!r}')
+        p = p / f'{name}.pt'
+        if not p.exists():
+            p = p.with_suffix('.pth')
         if not p.exists():
             return
         self.load_weights(model, str(p))

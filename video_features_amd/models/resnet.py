@@ -69,7 +69,7 @@ class Bottleneck(nn.Module):
                 and x.dtype == torch.bfloat16
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and ops.hip_available()
-                and not __import__('os').environ.get('VFA_NO_LTGEMM'))
+                and not ops._env_flag('VFA_NO_LTGEMM'))
 
     def forward(self, x):
         if self._fused_ready(x):

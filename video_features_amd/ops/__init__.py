@@ -22,6 +22,13 @@ _ext = None
 _ext_err: Optional[str] = None
 
 
+def _env_flag(name: str) -> bool:
+    """Boolean env knob: '', '0', 'false', 'no', 'off' (any case) are False —
+    so VFA_X=0 really disables (documented knob-table semantics)."""
+    return os.environ.get(name, '').strip().lower() not in (
+        '', '0', 'false', 'no', 'off')
+
+
 def _load_extension():
     global _ext, _ext_err
     if _ext is not None or _ext_err is not None:
@@ -41,7 +48,7 @@ def hip_available() -> bool:
 def _use_hip(t: torch.Tensor) -> bool:
     if not t.is_cuda:
         return False
-    if os.environ.get('VFA_FORCE_TORCH_OPS'):
+    if _env_flag('VFA_FORCE_TORCH_OPS'):
         return False
     ext = _load_extension()
     if ext is None:
@@ -217,7 +224,7 @@ def corr_lookup(pyramid, coords: torch.Tensor, radius: int = 4,
     if _use_hip(coords) and radius == 4:
         return _ext.corr_lookup(list(pyramid), coords.contiguous(), nhwc,
                                 out_dtype,
-                                bool(os.environ.get('VFA_CORR_GMEM')))
+                                _env_flag('VFA_CORR_GMEM'))
     # torch reference: per-level grid_sample of the displacement window
     b, _, h, w = coords.shape
     r = radius
@@ -361,7 +368,7 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
     n = weight.shape[0]
     if (_use_hip(x) and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16 and k % 8 == 0 and n >= 16
-            and not os.environ.get('VFA_NO_LTGEMM')):
+            and not _env_flag('VFA_NO_LTGEMM')):
         x2 = x.reshape(-1, k).contiguous()
         r2 = res.reshape(-1, n).contiguous() if res is not None else None
         out = _ext.linear_act(x2, weight.contiguous(), bias, r2,

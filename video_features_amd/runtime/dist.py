@@ -126,34 +126,49 @@ def _worker(rank: int, world: int, devices: List[str], cfg: Config,
         models = extractor.models_for(device)
         broadcast_models(models, src=0)
         if tp:
-            idxs = torch.arange(len(extractor.path_list),
-                                dtype=torch.long).to(device)
-        else:
-            idxs = shard_indices(len(extractor.path_list), rank,
-                                 world).to(device)
-        feats_list = extractor(idxs)
-        if tp:
+            # every rank runs EVERY video, owning its stride of the sliding
+            # windows; videos are processed ONE AT A TIME — each video's
+            # shards are gathered, merged and sunk on rank 0 before the next
+            # starts, so memory stays bounded by one video and --resume
+            # works (rank 0 checks existing outputs and all ranks agree)
             group = gather_group or (dist.new_group(backend='gloo')
                                      if backend == 'nccl' else dist.group.WORLD)
-            gathered = [None] * world if rank == 0 else None
-            dist.gather_object(feats_list, gathered, dst=0, group=group)
-            if rank == 0:
-                from .sinks import action_on_extraction
-                merged_all = [merge_temporal_shards(
-                    [g[i] for g in gathered],
-                    meta_keys=('fps',))
-                    for i in range(len(feats_list))]
-                for i, feats in enumerate(merged_all):
-                    if cfg.gather_features:
+            from .sinks import action_on_extraction
+            results = []
+            for i, video_path in enumerate(extractor.path_list):
+                if cfg.resume and not cfg.gather_features:
+                    skip = [extractor._already_done(video_path)
+                            if rank == 0 else None]
+                    dist.broadcast_object_list(skip, src=0, group=group)
+                    if skip[0]:
+                        extractor.progress.update()
                         continue
-                    action_on_extraction(
-                        feats, extractor._stem_path(extractor.path_list[i]),
-                        extractor.output_path, cfg.on_extraction,
-                        cfg.output_direct, cfg.feature_type)
-                if cfg.gather_features and return_dict is not None:
-                    return_dict['features'] = merged_all
+                feats = extractor(torch.tensor([i], dtype=torch.long,
+                                               device=device))
+                part = feats[0] if feats else None
+                gathered = [None] * world if rank == 0 else None
+                dist.gather_object(part, gathered, dst=0, group=group)
+                if rank == 0:
+                    if any(g is None for g in gathered):
+                        print(f'temporal-parallel: a rank failed on '
+                              f'{video_path}; skipping it')
+                        continue
+                    merged = merge_temporal_shards(gathered,
+                                                   meta_keys=('fps',))
+                    if cfg.gather_features:
+                        results.append(merged)
+                    else:
+                        action_on_extraction(
+                            merged, extractor._stem_path(video_path),
+                            extractor.output_path, cfg.on_extraction,
+                            cfg.output_direct, cfg.feature_type)
+            if rank == 0 and cfg.gather_features and return_dict is not None:
+                return_dict['features'] = results
             dist.barrier()
             return
+        idxs = shard_indices(len(extractor.path_list), rank,
+                             world).to(device)
+        feats_list = extractor(idxs)
         if cfg.gather_features:
             shard = [(int(i), f) for i, f in zip(idxs.tolist(), feats_list)]
             gathered: Optional[List] = [None] * world if rank == 0 else None

@@ -14,6 +14,7 @@ Scaling note: run one server process per GPU (HIP_VISIBLE_DEVICES=K) behind
 any HTTP load balancer — extraction is embarrassingly parallel across
 videos, so no cross-process state is needed.
 """
+import asyncio
 import io
 import os
 import tempfile
@@ -35,7 +36,10 @@ def create_app(cfg: Config):
                           else 'cuda:0')
     app = FastAPI(title='video-features-mi355x',
                   description=f'{cfg.feature_type} feature extraction')
-    state = {'extractor': None}
+    # the extractor is shared mutable state (ex.path_list is swapped per
+    # request): the lock serializes extractions, and run_in_executor keeps
+    # the blocking GPU work OFF the event loop so /health stays responsive
+    state = {'extractor': None, 'lock': asyncio.Lock()}
 
     def extractor(path: str):
         # built on first request (the input lister validates paths exist,
@@ -69,8 +73,12 @@ def create_app(cfg: Config):
             f.write(data)
             path = f.name
         try:
-            ex = extractor(path)
-            out = ex(torch.zeros(1, dtype=torch.long, device=device))
+            def run():
+                ex = extractor(path)
+                return ex(torch.zeros(1, dtype=torch.long, device=device))
+            async with state['lock']:
+                out = await asyncio.get_event_loop().run_in_executor(
+                    None, run)
             if not out:
                 raise HTTPException(422, 'extraction failed for this file')
             buf = io.BytesIO()

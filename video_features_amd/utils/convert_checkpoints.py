@@ -31,15 +31,21 @@ import torch
 
 def convert_clip_visual(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     """OpenAI CLIP (full model or visual tower) → VisionTransformer keys."""
+    # A full CLIP state dict carries the image encoder under 'visual.' and
+    # the text tower at top level (including 'transformer.resblocks.*', which
+    # would otherwise collide with the renamed visual blocks). If any
+    # 'visual.' key exists, keep ONLY the visual tower; the bare-key path
+    # below is for visual-tower-only dicts.
+    if any(k.startswith('visual.') for k in sd):
+        sd = {k: v for k, v in sd.items() if k.startswith('visual.')}
     out = {}
     for k, v in sd.items():
         if k.startswith('visual.'):
             k = k[len('visual.'):]
         elif any(k.startswith(p) for p in
-                 ('transformer.', 'token_embedding', 'text_projection',
-                  'logit_scale', 'ln_final', 'positional_embedding')) \
-                and not k.startswith('transformer.resblocks'):
-            # text-tower keys of a full CLIP model: not part of the image
+                 ('token_embedding', 'text_projection',
+                  'logit_scale', 'ln_final')):
+            # stray text-tower keys in a bare dict: not part of the image
             # encoder
             continue
         k = k.replace('transformer.resblocks.', 'blocks.')
