@@ -39,6 +39,30 @@ def test_r21d_shapes_and_params():
     assert f.shape == (1, 512)
 
 
+def test_r21d_34_depth_and_equivalence():
+    """R(2+1)D-34 (BASELINE config 5's depth): (3,4,6,3) basic blocks; the
+    flattened-time path must match the conv3d composition."""
+    from video_features_amd.models.r21d import R2Plus1D34
+    torch.manual_seed(1)
+    m = R2Plus1D34().eval()
+    assert len(m.layer1) == 3 and len(m.layer2) == 4
+    assert len(m.layer3) == 6 and len(m.layer4) == 3
+    x = torch.randn(1, 3, 8, 32, 32)
+
+    def ref(m, x):
+        x = m.stem(x)
+        for layer in (m.layer1, m.layer2, m.layer3, m.layer4):
+            for blk in layer:
+                x = blk.forward(x)
+        return m.avgpool(x).flatten(1)
+
+    with torch.no_grad():
+        a = m.forward_features(x)
+        b = ref(m, x)
+    assert a.shape == (1, 512)
+    torch.testing.assert_close(a, b, rtol=1e-4, atol=1e-5)
+
+
 def test_resnet_extractor_end_to_end(y4m_video):
     from video_features_amd.extractors.resnet import ExtractResNet
     cfg = Config(feature_type='resnet18', video_paths=[y4m_video],

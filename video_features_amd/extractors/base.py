@@ -31,6 +31,28 @@ from ..runtime.sinks import (action_on_extraction, make_output_path,
 from ..runtime.progress import make_progress
 
 
+def decode_ahead(read_chunk, chunk_specs):
+    """Iterate ``(spec, read_chunk(spec))`` with the NEXT chunk's decode
+    running on a worker thread while the caller runs the current chunk on
+    the GPU — intra-video decode/compute overlap for extractors that stream
+    a video in batches (ResNet/RAFT/PWC; the per-video ``prepare`` hook
+    covers the whole-video-decode families)."""
+    from concurrent.futures import ThreadPoolExecutor
+    chunk_specs = list(chunk_specs)
+    if len(chunk_specs) <= 1:
+        for spec in chunk_specs:
+            yield spec, read_chunk(spec)
+        return
+    with ThreadPoolExecutor(max_workers=1,
+                            thread_name_prefix='vfa-chunk-decode') as pool:
+        fut = pool.submit(read_chunk, chunk_specs[0])
+        for i, spec in enumerate(chunk_specs):
+            cur = fut.result()
+            if i + 1 < len(chunk_specs):
+                fut = pool.submit(read_chunk, chunk_specs[i + 1])
+            yield spec, cur
+
+
 class BaseExtractor(torch.nn.Module):
     feature_type: str = ''
 

@@ -14,7 +14,7 @@ import torch
 from .. import transforms as T
 from ..io.video import open_video
 from ..models.pwc import PWCNet
-from .base import BaseExtractor
+from .base import BaseExtractor, decode_ahead
 
 
 class ExtractPWC(BaseExtractor):
@@ -43,10 +43,20 @@ class ExtractPWC(BaseExtractor):
         n = reader.frame_count
         dtype = self.compute_dtype(device)
         flows = []
-        start = 0
+        # (start, stop) windows with 1-frame carry-over; next window decodes
+        # ahead on a worker thread (see raft extractor)
+        spans, start = [], 0
         while start < n - 1:
             stop = min(start + self.batch_size, n)
-            frames = torch.from_numpy(reader.read_frames(range(start, stop)))
+            spans.append((start, stop))
+            start = stop - 1
+        pin = device.type == 'cuda'
+
+        def read_chunk(span):
+            t = torch.from_numpy(reader.read_frames(range(*span)))
+            return t.pin_memory() if pin else t
+
+        for (start, stop), frames in decode_ahead(read_chunk, spans):
             batch = self._prep(frames).to(device=device, dtype=dtype,
                                           non_blocking=True)
             flow = model(batch[:-1], batch[1:])
@@ -63,7 +73,6 @@ class ExtractPWC(BaseExtractor):
                     flow[0].float().cpu().numpy().transpose(1, 2, 0))
                 save_ppm(_os.path.join(
                     self.tmp_path, f'flow_vis_{start:05d}.ppm'), img)
-            start = stop - 1
         features = torch.cat(flows).numpy() if flows else np.zeros((0, 2, 0, 0))
         return {
             self.feature_type: features,

@@ -17,7 +17,7 @@ import torch
 from .. import transforms as T
 from ..io.video import open_video
 from ..models.raft import RAFT, InputPadder
-from .base import BaseExtractor
+from .base import BaseExtractor, decode_ahead
 
 
 class ExtractRAFT(BaseExtractor):
@@ -46,10 +46,21 @@ class ExtractRAFT(BaseExtractor):
         n = reader.frame_count
         dtype = self.compute_dtype(device)
         flows = []
-        start = 0
+        # (start, stop) windows with the reference's 1-frame carry-over
+        # (extract_raft.py:143-146); the next window's decode runs ahead on
+        # a worker thread while this window's flow runs on the GPU
+        spans, start = [], 0
         while start < n - 1:
             stop = min(start + self.batch_size, n)
-            frames = torch.from_numpy(reader.read_frames(range(start, stop)))
+            spans.append((start, stop))
+            start = stop - 1
+        pin = device.type == 'cuda'
+
+        def read_chunk(span):
+            t = torch.from_numpy(reader.read_frames(range(*span)))
+            return t.pin_memory() if pin else t
+
+        for (start, stop), frames in decode_ahead(read_chunk, spans):
             batch = self._prep(frames).to(device=device, dtype=dtype,
                                           non_blocking=True)
             padder = InputPadder(batch.shape)
@@ -71,7 +82,6 @@ class ExtractRAFT(BaseExtractor):
                     .transpose(1, 2, 0))
                 save_ppm(_os.path.join(
                     self.tmp_path, f'flow_vis_{start:05d}.ppm'), img)
-            start = stop - 1   # carry the last frame over for continuity
         features = torch.cat(flows).numpy() if flows else np.zeros((0, 2, 0, 0))
         return {
             self.feature_type: features,

@@ -16,7 +16,7 @@ from .. import transforms as T
 from ..io.sampling import timestamps_ms
 from ..io.video import open_video
 from ..models.resnet import build_resnet
-from .base import BaseExtractor
+from .base import BaseExtractor, decode_ahead
 
 
 class ExtractResNet(BaseExtractor):
@@ -37,9 +37,15 @@ class ExtractResNet(BaseExtractor):
         dtype = self.compute_dtype(device)
         feats, idx_done = [], []
         n = reader.frame_count
-        for start in range(0, n, self.batch_size):
-            idxs = list(range(start, min(start + self.batch_size, n)))
-            frames_u8 = torch.from_numpy(reader.read_frames(idxs))
+        pin = device.type == 'cuda'
+
+        def read_chunk(idxs):
+            t = torch.from_numpy(reader.read_frames(idxs))
+            return t.pin_memory() if pin else t
+
+        chunks = [list(range(s, min(s + self.batch_size, n)))
+                  for s in range(0, n, self.batch_size)]
+        for idxs, frames_u8 in decode_ahead(read_chunk, chunks):
             if device.type == 'cuda':   # preprocess on the GPU (u8 upload)
                 frames_u8 = frames_u8.to(device, non_blocking=True)
             batch = T.imagenet_preprocess(frames_u8).to(dtype)

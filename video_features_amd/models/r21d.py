@@ -1,4 +1,4 @@
-"""Native R(2+1)D-18 video network.
+"""Native R(2+1)D video networks (depths 18 and 34).
 
 The reference uses ``torchvision.models.video.r2plus1d_18``
 (reference models/r21d/extract_r21d.py:57).  From-scratch implementation of
@@ -123,7 +123,12 @@ class R21DBlock(nn.Module):
 
 
 class R2Plus1D18(nn.Module):
-    def __init__(self, num_classes: int = 400):
+    """R(2+1)D with basic blocks.  ``layers`` gives the per-stage block
+    counts: (2,2,2,2) is R(2+1)D-18 (torchvision's ``r2plus1d_18``,
+    reference extract_r21d.py:57), (3,4,6,3) is R(2+1)D-34 (the IG-65M
+    depth BASELINE.json config 5 names)."""
+
+    def __init__(self, num_classes: int = 400, layers=(2, 2, 2, 2)):
         super().__init__()
         # R(2+1)D stem: (1,7,7) spatial s(1,2,2) into 45 ch, then (3,1,1)
         self.stem = nn.Sequential(
@@ -131,10 +136,16 @@ class R2Plus1D18(nn.Module):
             nn.BatchNorm3d(45), nn.ReLU(inplace=True),
             nn.Conv3d(45, 64, (3, 1, 1), (1, 1, 1), (1, 0, 0), bias=False),
             nn.BatchNorm3d(64), nn.ReLU(inplace=True))
-        self.layer1 = nn.Sequential(R21DBlock(64, 64), R21DBlock(64, 64))
-        self.layer2 = nn.Sequential(R21DBlock(64, 128, 2), R21DBlock(128, 128))
-        self.layer3 = nn.Sequential(R21DBlock(128, 256, 2), R21DBlock(256, 256))
-        self.layer4 = nn.Sequential(R21DBlock(256, 512, 2), R21DBlock(512, 512))
+
+        def stage(in_p, out_p, n, stride):
+            blocks = [R21DBlock(in_p, out_p, stride)]
+            blocks += [R21DBlock(out_p, out_p) for _ in range(n - 1)]
+            return nn.Sequential(*blocks)
+
+        self.layer1 = stage(64, 64, layers[0], 1)
+        self.layer2 = stage(64, 128, layers[1], 2)
+        self.layer3 = stage(128, 256, layers[2], 2)
+        self.layer4 = stage(256, 512, layers[3], 2)
         self.avgpool = nn.AdaptiveAvgPool3d(1)
         self.feat_dim = 512
         self.fc = nn.Linear(512, num_classes)
@@ -180,3 +191,8 @@ class R2Plus1D18(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         return self.fc(self.forward_features(x))
+
+
+def R2Plus1D34(num_classes: int = 400) -> R2Plus1D18:
+    """R(2+1)D-34: basic blocks at depths (3, 4, 6, 3)."""
+    return R2Plus1D18(num_classes, layers=(3, 4, 6, 3))
