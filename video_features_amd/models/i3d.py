@@ -73,7 +73,12 @@ class Unit3D(nn.Module):
         pad = tuple(k // 2 for k in self.kernel) if self.static_same else 0
         self.conv = nn.Conv3d(in_ch, out_ch, self.kernel, self.stride,
                               padding=pad, bias=use_bias)
-        self.bn = nn.BatchNorm3d(out_ch, eps=1e-3) if use_bn else None
+        # torch-default eps (1e-5): the reference's Unit3Dpy uses
+        # BatchNorm3d defaults (i3d_net.py:92), and published i3d_rgb.pt /
+        # i3d_flow.pt checkpoints are calibrated against that — 1e-3 here
+        # cost ~1e-3 absolute feature divergence (caught by
+        # tests/test_reference_parity.py)
+        self.bn = nn.BatchNorm3d(out_ch) if use_bn else None
         self.activation = activation
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:

@@ -61,7 +61,7 @@ class Decoder(nn.Module):
 
     DENSE = [128, 128, 96, 64, 32]
 
-    def __init__(self, feat_ch: int, top: bool = False):
+    def __init__(self, feat_ch: int, top: bool = False, last: bool = False):
         super().__init__()
         self.top = top
         in_ch = 81 if top else 81 + feat_ch + 2 + 2   # corr + f1 + upflow + upfeat
@@ -73,8 +73,13 @@ class Decoder(nn.Module):
         self.convs = nn.ModuleList(convs)
         self.predict = nn.Conv2d(c, 2, 3, 1, 1)
         self.out_channels = c
-        self.upflow = nn.ConvTranspose2d(2, 2, 4, 2, 1)
-        self.upfeat = nn.ConvTranspose2d(c, 2, 4, 2, 1)
+        # upflow/upfeat feed the NEXT (finer) level; the bottom decoder
+        # (level 2) has no consumer, so the parameter set matches the
+        # reference's exactly (its Decoder(l) holds the upsamplers the
+        # consumer side, pwc_net.py:119-120 — same tensors, shifted owner)
+        if not last:
+            self.upflow = nn.ConvTranspose2d(2, 2, 4, 2, 1)
+            self.upfeat = nn.ConvTranspose2d(c, 2, 4, 2, 1)
 
     def forward(self, f1: torch.Tensor, f2: torch.Tensor,
                 upflow: Optional[torch.Tensor],
@@ -121,7 +126,7 @@ class PWCNet(nn.Module):
         self.decoder5 = Decoder(ch[4])
         self.decoder4 = Decoder(ch[3])
         self.decoder3 = Decoder(ch[2])
-        self.decoder2 = Decoder(ch[1])
+        self.decoder2 = Decoder(ch[1], last=True)
         self.refiner = Refiner(self.decoder2.out_channels)
 
     def forward(self, im1: torch.Tensor, im2: torch.Tensor) -> torch.Tensor:

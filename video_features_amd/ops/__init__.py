@@ -233,8 +233,10 @@ def corr_lookup(pyramid, coords: torch.Tensor, radius: int = 4,
     for lvl, corr in enumerate(pyramid):
         dx = torch.linspace(-r, r, 2 * r + 1, device=coords.device,
                             dtype=torch.float32)
-        delta = torch.stack(torch.meshgrid(dx, dx, indexing='ij'),
-                            dim=-1).flip(-1)
+        # channel t = i*9+j offsets (x + d_i, y + d_j) — the REFERENCE's
+        # order (corr.py:39 stacks meshgrid(dy, dx) last and adds it to
+        # (x, y) coords), which pretrained motion-encoder weights consume
+        delta = torch.stack(torch.meshgrid(dx, dx, indexing='ij'), dim=-1)
         centroid = cc.reshape(b * h * w, 1, 1, 2) / (2 ** lvl)
         window = centroid + delta[None]
         sampled = grid_sample_bilinear(corr, window)

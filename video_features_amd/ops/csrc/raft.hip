@@ -73,7 +73,9 @@ __global__ void corr_lookup_lds_kernel(const float* __restrict__ l0,
   const int ctot = levels * KK;
   for (int c = tid; c < ctot; c += 64) {
     const int l = c / KK, t = c % KK;
-    const int dy = t / K - RAD, dx = t % K - RAD;
+    const int dx = t / K - RAD, dy = t % K - RAD;  // tap t = i*9+j: i
+    // offsets X, j offsets Y — the reference's channel order
+    // (corr.py:39 stacks (dy,dx) last, added to (x,y) coords)
     const float inv = 1.0f / (float)(1 << l);
     const float sx = cx * inv + dx, sy = cy * inv + dy;
     const int h = lh[l], w = lw[l];
@@ -122,7 +124,9 @@ __global__ void corr_lookup_gmem_kernel(const float* __restrict__ l0,
     const float cx = coords[(b * 2 + 0) * hw + p];
     const float cy = coords[(b * 2 + 1) * hw + p];
     const int l = c / KK, t = c % KK;
-    const int dy = t / K - RAD, dx = t % K - RAD;
+    const int dx = t / K - RAD, dy = t % K - RAD;  // tap t = i*9+j: i
+    // offsets X, j offsets Y — the reference's channel order
+    // (corr.py:39 stacks (dy,dx) last, added to (x,y) coords)
     const float inv = 1.0f / (float)(1 << l);
     const float sx = cx * inv + dx, sy = cy * inv + dy;
     const int h = lh[l], w = lw[l];

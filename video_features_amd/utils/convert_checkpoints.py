@@ -87,6 +87,47 @@ def convert_r21d(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     return out
 
 
+_PWC_WORD = {'One': 1, 'Two': 2, 'Thr': 3, 'Fou': 4, 'Fiv': 5, 'Six': 6}
+
+
+def convert_pwc(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
+    """Reference ``pwc_net_sintel.pt`` (sniklaus ``module*`` scheme,
+    reference models/pwc/pwc_src/pwc_net.py) → this package's PWCNet keys.
+
+    The reference's ``Decoder(l)`` owns the upsamplers applied to level
+    ``l+1``'s output (pwc_net.py:119-120); here they live with the
+    producing decoder, so ``module{l}.moduleUpflow`` maps to
+    ``decoder{l+1}.upflow``.
+    """
+    out = {}
+    word = '|'.join(_PWC_WORD)
+    for k, v in sd.items():
+        m = re.match(rf'moduleExtractor\.module({word})\.(\d+)\.(.*)', k)
+        if m:
+            lvl, i, p = _PWC_WORD[m.group(1)], int(m.group(2)), m.group(3)
+            out[f'extractor.levels.{lvl - 1}.{i // 2}.0.{p}'] = v
+            continue
+        m = re.match(rf'moduleRefiner\.moduleMain\.(\d+)\.(.*)', k)
+        if m:
+            i, p = int(m.group(1)), m.group(2)
+            out[f'refiner.net.6.{p}' if i == 12
+                else f'refiner.net.{i // 2}.0.{p}'] = v
+            continue
+        m = re.match(rf'module({word})\.module({word})\.0\.(.*)', k)
+        if m:
+            lvl, c, p = _PWC_WORD[m.group(1)], _PWC_WORD[m.group(2)], m.group(3)
+            out[f'decoder{lvl}.predict.{p}' if c == 6
+                else f'decoder{lvl}.convs.{c - 1}.0.{p}'] = v
+            continue
+        m = re.match(rf'module({word})\.moduleUp(flow|feat)\.(.*)', k)
+        if m:
+            lvl, kind, p = _PWC_WORD[m.group(1)], m.group(2), m.group(3)
+            out[f'decoder{lvl + 1}.up{kind}.{p}'] = v
+            continue
+        out[k] = v
+    return out
+
+
 def convert_vggish(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     """harritaylor/torchvggish keys → this package's VGGish keys."""
     return {('net.' + k if not k.startswith('net.') else k): v
@@ -100,6 +141,8 @@ def convert_auto(sd: Dict[str, torch.Tensor]) -> Dict[str, torch.Tensor]:
     keys = list(sd.keys())
     if any('.conv3d.' in k or '.batch3d.' in k for k in keys):
         return convert_i3d(sd)
+    if any(k.startswith('moduleExtractor.') for k in keys):
+        return convert_pwc(sd)
     if any('in_proj_weight' in k or k.startswith('visual.') for k in keys):
         return convert_clip_visual(sd)
     if any(re.match(r'layer\d\.\d\.conv\d\.0\.0\.', k) for k in keys):
