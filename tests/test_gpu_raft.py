@@ -32,9 +32,11 @@ def _ref_corr_lookup(pyramid, coords, radius=4):
     cc = coords.permute(0, 2, 3, 1)
     out = []
     for lvl, corr in enumerate(pyramid):
+        # tap t = i*9+j offsets (x + d_i, y + d_j) — the reference's
+        # channel order (corr.py:39 adds its (dy,dx)-stacked meshgrid to
+        # (x,y) coords); see tests/test_reference_parity.py
         dx = torch.linspace(-r, r, 2 * r + 1, device=coords.device)
-        delta = torch.stack(torch.meshgrid(dx, dx, indexing='ij'),
-                            dim=-1).flip(-1)
+        delta = torch.stack(torch.meshgrid(dx, dx, indexing='ij'), dim=-1)
         centroid = cc.reshape(b * h * w, 1, 1, 2) / (2 ** lvl)
         window = (centroid + delta[None])
         gh, gw = corr.shape[-2:]
