@@ -355,7 +355,8 @@ def maxpool2d_same(x: torch.Tensor, kernel, stride,
     return torch.nn.functional.max_pool2d(xp, tuple(kernel), tuple(stride))
 
 
-_ACT_IDS = {'none': 0, 'relu': 1, 'quick_gelu': 2, 'gelu': 3}
+_ACT_IDS = {'none': 0, 'relu': 1, 'quick_gelu': 2, 'gelu': 3,
+            'leaky_relu': 4}
 
 
 def linear_act(x: torch.Tensor, weight: torch.Tensor,
@@ -412,6 +413,45 @@ def conv1x1_act(x: torch.Tensor, weight: torch.Tensor,
         if res is not None else None
     y = linear_act(x2, weight.reshape(weight.shape[0], cin), bias, act, r2)
     return y.reshape(b, h, w, weight.shape[0]).permute(0, 3, 1, 2)
+
+
+def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
+               bias: Optional[torch.Tensor] = None,
+               stride=1, padding=0, act: str = 'none',
+               res: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Fused conv2d + bias + activation (+ residual) on channels_last bf16.
+
+    GPU path: the in-tree implicit-GEMM MFMA kernel (conv2d.hip) — the
+    hand-written CDNA4 replacement for MIOpen's igemm on the 3x3 / 1x5 /
+    5x1 conv shapes of the ResNet / RAFT / I3D / VGGish stacks (reference
+    conv stacks at models/raft/raft_src/extractor.py:118-192,
+    models/i3d/i3d_src/i3d_net.py:37-105).  CPU / unsupported shapes:
+    F.conv2d composition.
+    """
+    sh, sw = (stride, stride) if isinstance(stride, int) else stride
+    ph, pw = (padding, padding) if isinstance(padding, int) else padding
+    if (_use_hip(x) and x.dtype == torch.bfloat16
+            and weight.dtype == torch.bfloat16
+            and x.shape[1] % 8 == 0
+            and x.is_contiguous(memory_format=torch.channels_last)
+            and not _env_flag('VFA_NO_CONV')):
+        w_cl = weight.contiguous(memory_format=torch.channels_last)
+        r = res.contiguous(memory_format=torch.channels_last) \
+            if res is not None else None
+        return _ext.conv2d_nhwc(x, w_cl, bias, r, sh, sw, ph, pw,
+                                _ACT_IDS[act])
+    y = torch.nn.functional.conv2d(x, weight, bias, (sh, sw), (ph, pw))
+    if res is not None:
+        y = y + res
+    if act == 'relu':
+        y = torch.nn.functional.relu(y)
+    elif act == 'quick_gelu':
+        y = y * torch.sigmoid(1.702 * y)
+    elif act == 'gelu':
+        y = torch.nn.functional.gelu(y, approximate='tanh')
+    elif act == 'leaky_relu':
+        y = torch.nn.functional.leaky_relu(y, 0.1)
+    return y
 
 
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:

@@ -48,6 +48,11 @@ void vfa_linear_act(const void*, const void*, const void*, const void*,
                     void*, int, int, int, int, hipStream_t);
 void vfa_temporal_merge(const void*, void*, int, int, int, int, int, int,
                         int, long long, int, int, hipStream_t);
+void vfa_conv2d_nhwc(const void*, const void*, const void*, const void*,
+                     void*, int, int, int, int, int, int, int, int, int,
+                     int, int, int, hipStream_t);
+void vfa_pad2d_nhwc(const void*, void*, int, int, int, int, int, int, int,
+                    int, hipStream_t);
 }
 
 namespace {
@@ -412,6 +417,57 @@ torch::Tensor linear_act(torch::Tensor x, torch::Tensor w,
   return out;
 }
 
+torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
+                          c10::optional<torch::Tensor> bias,
+                          c10::optional<torch::Tensor> res,
+                          int64_t stride_h, int64_t stride_w,
+                          int64_t pad_h, int64_t pad_w, int64_t act) {
+  // x (B, C, H, W) channels_last bf16; w (K, C, KH, KW) channels_last bf16
+  // (physical (K, KH, KW, C) = the (N, Kr) B-operand); out (B, K, OH, OW)
+  // channels_last.  Implicit-GEMM MFMA kernel; input pre-padded by the
+  // pad2d kernel when pad > 0.
+  TORCH_CHECK(x.is_cuda() && x.dim() == 4 && cl_contig(x),
+              "x must be channels_last");
+  TORCH_CHECK(w.dim() == 4 && cl_contig(w), "w must be channels_last");
+  TORCH_CHECK(x.scalar_type() == torch::kBFloat16 &&
+              w.scalar_type() == torch::kBFloat16);
+  const int b = (int)x.size(0), c = (int)x.size(1);
+  const int h = (int)x.size(2), ww = (int)x.size(3);
+  const int kout = (int)w.size(0), kh = (int)w.size(2), kw = (int)w.size(3);
+  TORCH_CHECK(w.size(1) == c && c % 8 == 0, "C % 8 == 0 required");
+  const int hp = h + 2 * (int)pad_h, wp = ww + 2 * (int)pad_w;
+  const int oh = (hp - kh) / (int)stride_h + 1;
+  const int ow = (wp - kw) / (int)stride_w + 1;
+  auto stream = current_stream();
+  torch::Tensor xp = x;
+  if (pad_h > 0 || pad_w > 0) {
+    xp = torch::empty({b, hp, wp, c}, x.options());
+    vfa_pad2d_nhwc(x.data_ptr(), xp.data_ptr(), b, h, ww, c, (int)pad_h,
+                   (int)pad_h, (int)pad_w, (int)pad_w, stream);
+    xp = xp.permute({0, 3, 1, 2});   // logical NCHW view, CL physical
+  }
+  const void* bptr = nullptr;
+  torch::Tensor bc;
+  if (bias.has_value()) {
+    bc = bias->contiguous().to(torch::kBFloat16);
+    TORCH_CHECK(bc.numel() == kout);
+    bptr = bc.data_ptr();
+  }
+  const void* rptr = nullptr;
+  if (res.has_value()) {
+    TORCH_CHECK(cl_contig(*res) && res->scalar_type() == torch::kBFloat16 &&
+                res->numel() == (long long)b * oh * ow * kout,
+                "residual must be CL bf16 of the output shape");
+    rptr = res->data_ptr();
+  }
+  auto out = torch::empty({(long)b, oh, ow, kout}, x.options())
+                 .permute({0, 3, 1, 2});
+  vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr, out.data_ptr(),
+                  b, hp, wp, c, oh, ow, kout, kh, kw, (int)stride_h,
+                  (int)stride_w, (int)act, stream);
+  return out;
+}
+
 torch::Tensor temporal_merge(torch::Tensor y, int64_t b, int64_t kt,
                              int64_t st, int64_t p0, bool relu) {
   // y (B*T, kt*O, H, W) channels_last -> (B*T', O, H, W) channels_last
@@ -451,6 +507,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("maxpool3d_same", &maxpool3d_same);
   m.def("maxpool2d_same", &maxpool2d_same);
   m.def("linear_act", &linear_act);
+  m.def("conv2d_nhwc", &conv2d_nhwc);
   m.def("temporal_merge", &temporal_merge);
   m.attr("gfx_arch") = "gfx950";
 }

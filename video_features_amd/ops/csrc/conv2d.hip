@@ -1,0 +1,372 @@
+// Implicit-GEMM conv2d for gfx950 (CDNA4), NHWC bf16 — hand-written MFMA
+// replacement for MIOpen's igemm on the 3x3 / 1x5 / 5x1 / 7x7 hot shapes
+// of the ResNet / RAFT / I3D(flattened) / VGGish conv stacks (reference
+// conv stacks: models/resnet & models/raft/raft_src/extractor.py:118-192,
+// models/i3d/i3d_src/i3d_net.py:37-105, vggish_src/vggish.py:108-118).
+//
+// GEMM view:  C[M, N] = act( A[M, Kr] @ W[N, Kr]^T + bias [+ res] )
+//   M  = B*OH*OW output pixels, N = K_out channels,
+//   Kr = KH*KW*C with reduction index k = (r*KW + s)*C + c.
+// A is the im2col view of the PRE-PADDED input (B, Hp, Wp, C) — never
+// materialized: each staged A row is an input pixel's C-slice at tap
+// (r, s), address  base(m) + (r*Wp + s)*C + c  with
+// base(m) = ((b*Hp + oy*sh)*Wp + ox*sw)*C precomputed PER THREAD before
+// the K-loop (each thread always stages the same tile rows; only the
+// k-offset changes per K-tile).  W is torch's channels_last conv weight —
+// physically (K_out, KH, KW, C) = exactly the (N, Kr) row-major B operand,
+// so weights need NO reshuffling and B staging is identical to the linear
+// kernel's.
+//
+// Machinery shared with gemm.hip (cdna_hip_programming.md §5): 256x256 /
+// 128x128 tiles, BK=64, double-buffered LDS staged by
+// __builtin_amdgcn_global_load_lds width 16 with the st_16x32 XOR swizzle
+// on the per-lane *source* address (LDS image stays lane-linear), glds for
+// the next K-tile spread across the current tile's two MFMA half-steps,
+// mfma_f32_16x16x32_bf16, fused bias+activation(+residual) epilogue.
+// Requires C % 8 == 0 (each lane's 16-B segment stays inside one tap).
+#include "vfa_common.h"
+
+typedef __bf16 bf16x8c __attribute__((ext_vector_type(8)));
+typedef float f32x4c __attribute__((ext_vector_type(4)));
+
+namespace {
+
+constexpr int BK = 64;
+
+__device__ __forceinline__ int cswz(int byte_off) {
+  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+}
+
+__device__ __forceinline__ float conv_act_f(float x, int kind) {
+  if (kind == 1) return fmaxf(x, 0.f);
+  if (kind == 2) return x / (1.0f + __expf(-1.702f * x));  // QuickGELU
+  if (kind == 3) {
+    const float k0 = 0.7978845608028654f, k1 = 0.044715f;
+    return 0.5f * x * (1.0f + tanhf(k0 * (x + k1 * x * x * x)));
+  }
+  if (kind == 4) return fmaxf(x, 0.f) + 0.1f * fminf(x, 0.f);  // LeakyReLU .1
+  return x;
+}
+
+struct ConvGeom {
+  int b, hp, wp, cin;       // padded input
+  int oh, ow, kout;         // output
+  int kh, kw, sh, sw;       // filter / stride
+  int kr;                   // KH*KW*C
+};
+
+// decompose output-pixel index m -> flat element base into x_pad
+__device__ __forceinline__ long long pix_base(const ConvGeom& g, int m) {
+  const int ox = m % g.ow;
+  const int t = m / g.ow;
+  const int oy = t % g.oh;
+  const int bi = t / g.oh;
+  return (((long long)bi * g.hp + (long long)oy * g.sh) * g.wp +
+          (long long)ox * g.sw) * g.cin;
+}
+
+// k (reduction index) -> element offset into x_pad relative to pix_base
+__device__ __forceinline__ long long tap_off(const ConvGeom& g, int k) {
+  const int tap = k / g.cin;            // uniform-cost u32 div
+  const int c = k - tap * g.cin;
+  const int r = tap / g.kw;
+  const int s = tap - r * g.kw;
+  return (long long)(r * g.wp + s) * g.cin + c;
+}
+
+// Stage the (ROWS x 64) A-tile via glds: per-thread row bases precomputed
+// in abase[]; the per-lane k position adds the tap offset for this K-tile.
+// piece splits the wave's glds issues across MFMA half-steps (-1 = all).
+template <int ROWS, int WAVES>
+__device__ __forceinline__ void conv_stage_a(
+    const __bf16* __restrict__ x, const ConvGeom& g,
+    const long long* __restrict__ abase, int k0, char* lds_base, int wave,
+    int lane, int kfrac, int piece, int npieces) {
+  constexpr int NSUB = ROWS * 128 / 1024;
+  constexpr int PER_WAVE = NSUB / WAVES;
+#pragma unroll
+  for (int i = 0; i < PER_WAVE; ++i) {
+    if (piece >= 0 && (i * npieces) / PER_WAVE != piece) continue;
+    const int sub = wave * PER_WAVE + i;
+    const long long src_elem = abase[i] + tap_off(g, k0 + kfrac);
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int*>(x + src_elem),
+        reinterpret_cast<unsigned int*>(lds_base + sub * 1024), 16, 0, 0);
+  }
+}
+
+// Weight staging == linear kernel's stage_glds (W is (N, Kr) row-major).
+template <int ROWS, int WAVES>
+__device__ __forceinline__ void conv_stage_w(
+    const __bf16* __restrict__ wgt, long long row_stride, char* lds_base,
+    int wave, int lane, int piece, int npieces) {
+  constexpr int NSUB = ROWS * 128 / 1024;
+  constexpr int PER_WAVE = NSUB / WAVES;
+  const int off = lane * 16;
+  const int off_log = off ^ (((off >> 9) & 1) << 5);
+  const int r_in = off_log >> 7;
+  const int b_in = off_log & 127;
+#pragma unroll
+  for (int i = 0; i < PER_WAVE; ++i) {
+    if (piece >= 0 && (i * npieces) / PER_WAVE != piece) continue;
+    const int sub = wave * PER_WAVE + i;
+    const __bf16* src = wgt + (long long)(sub * 8 + r_in) * row_stride;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int*>(
+            reinterpret_cast<const char*>(src) + b_in),
+        reinterpret_cast<unsigned int*>(lds_base + sub * 1024), 16, 0, 0);
+  }
+}
+
+// Bounds-checked A staging (M-edge tiles and the ragged last K-tile):
+// same LDS image, zero fill.
+template <int ROWS, int THREADS>
+__device__ __forceinline__ void conv_guard_a(
+    const __bf16* __restrict__ x, const ConvGeom& g, int m0, int mtotal,
+    int k0, char* lds_base, int tid) {
+  for (int t = tid; t < ROWS * 8; t += THREADS) {  // 8 16-B segs per row
+    const int row = t >> 3, seg = t & 7;
+    uint4 v = {0u, 0u, 0u, 0u};
+    const int m = m0 + row;
+    const int k = k0 + seg * 8;
+    if (m < mtotal && k < g.kr) {
+      const long long src = pix_base(g, m) + tap_off(g, k);
+      v = *reinterpret_cast<const uint4*>(x + src);
+    }
+    *reinterpret_cast<uint4*>(lds_base + cswz(row * 128 + seg * 16)) = v;
+  }
+}
+
+template <int ROWS, int THREADS>
+__device__ __forceinline__ void conv_guard_w(
+    const __bf16* __restrict__ wgt, long long row_stride, int n0, int ntotal,
+    int k0, int kr, char* lds_base, int tid) {
+  for (int t = tid; t < ROWS * 8; t += THREADS) {
+    const int row = t >> 3, seg = t & 7;
+    uint4 v = {0u, 0u, 0u, 0u};
+    if (n0 + row < ntotal && k0 + seg * 8 < kr) {
+      v = *reinterpret_cast<const uint4*>(wgt + (long long)row * row_stride +
+                                          seg * 8);
+    }
+    *reinterpret_cast<uint4*>(lds_base + cswz(row * 128 + seg * 16)) = v;
+  }
+}
+
+// BIG: 256x256 tile, 8 waves (2Mx4N); else 128x128, 4 waves (2x2).
+template <int ACT, bool BIG>
+__global__ __launch_bounds__(BIG ? 512 : 256)
+void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
+                        const __bf16* __restrict__ wgt,
+                        const __bf16* __restrict__ bias,
+                        const __bf16* __restrict__ res,
+                        __bf16* __restrict__ out, ConvGeom g, int mtotal,
+                        int tiles_m, int tiles_n) {
+  constexpr int BM = BIG ? 256 : 128, BN = BIG ? 256 : 128;
+  constexpr int WAVES = BIG ? 8 : 4;
+  constexpr int WN = BIG ? 4 : 2;
+  constexpr int MI = BIG ? 8 : 4;
+  constexpr int NJ = 4;
+  constexpr int TILE_A = BM * BK * 2, TILE_B = BN * BK * 2;
+  constexpr int THREADS = BIG ? 512 : 256;
+  constexpr int PER_WAVE = (BM * 128 / 1024) / WAVES;
+
+  extern __shared__ __attribute__((aligned(1024))) char smem[];
+  auto sA = [&](int buf) { return smem + buf * (TILE_A + TILE_B); };
+  auto sB = [&](int buf) { return smem + buf * (TILE_A + TILE_B) + TILE_A; };
+
+  const int nwg = tiles_m * tiles_n;
+  int wg = blockIdx.x;
+  {
+    const int xcd = wg % 8, orig = wg / 8;
+    const int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+  }
+  const int tile_n = wg / tiles_m, tile_m = wg % tiles_m;
+  const int m0 = tile_m * BM, n0 = tile_n * BN;
+
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  const int lo = lane & 15, hi4 = lane >> 4;
+  const int wm = wave / WN, wn = wave % WN;
+
+  const int kr = g.kr, n = g.kout;
+  const int nk = (kr + BK - 1) / BK;
+  const int valid_m = mtotal - m0, valid_n = n - n0;
+  const bool tile_full = valid_m >= BM && valid_n >= BN;
+  const bool kfull = (kr % BK) == 0;
+
+  // per-thread A row bases for the glds path (same rows every K-tile);
+  // the per-lane swizzled byte offset contributes kfrac elements
+  const int off_log = (lane * 16) ^ ((((lane * 16) >> 9) & 1) << 5);
+  const int a_rin = off_log >> 7;
+  const int kfrac = (off_log & 127) >> 1;
+  long long abase[PER_WAVE];
+#pragma unroll
+  for (int i = 0; i < PER_WAVE; ++i) {
+    const int sub = wave * PER_WAVE + i;
+    const int m = m0 + sub * 8 + a_rin;
+    abase[i] = pix_base(g, m < mtotal ? m : mtotal - 1);
+  }
+
+  auto stage = [&](int buf, int kt) {
+    const int k0 = kt * BK;
+    if (tile_full && (kfull || kt + 1 < nk)) {
+      conv_stage_a<BM, WAVES>(x, g, abase, k0, sA(buf), wave, lane, kfrac,
+                              -1, 1);
+      conv_stage_w<BN, WAVES>(wgt + (long long)n0 * kr + k0, kr, sB(buf),
+                              wave, lane, -1, 1);
+    } else {
+      conv_guard_a<BM, THREADS>(x, g, m0, mtotal, k0, sA(buf), threadIdx.x);
+      conv_guard_w<BN, THREADS>(wgt + (long long)n0 * kr + k0, kr, n0, n,
+                                k0, kr, sB(buf), threadIdx.x);
+    }
+  };
+
+  stage(0, 0);
+
+  f32x4c acc[MI][NJ];
+#pragma unroll
+  for (int i = 0; i < MI; ++i)
+#pragma unroll
+    for (int j = 0; j < NJ; ++j) acc[i][j] = f32x4c{0.f, 0.f, 0.f, 0.f};
+
+  for (int kt = 0; kt < nk; ++kt) {
+    __syncthreads();
+    const int cur = kt & 1;
+    const int k0_nxt = (kt + 1) * BK;
+    const bool glds_nxt = (kt + 1 < nk) && tile_full &&
+                          (kfull || kt + 2 < nk);
+    if ((kt + 1 < nk) && !glds_nxt) stage(1 - cur, kt + 1);
+#pragma unroll
+    for (int kk = 0; kk < 2; ++kk) {
+      bf16x8c afr[MI], bfr[NJ];
+#pragma unroll
+      for (int j = 0; j < NJ; ++j) {
+        const int brow = wn * (NJ * 16) + j * 16 + lo;
+        bfr[j] = *reinterpret_cast<const bf16x8c*>(
+            sB(cur) + cswz(brow * 128 + kk * 64 + hi4 * 16));
+      }
+#pragma unroll
+      for (int i = 0; i < MI; ++i) {
+        const int arow = wm * (MI * 16) + i * 16 + lo;
+        afr[i] = *reinterpret_cast<const bf16x8c*>(
+            sA(cur) + cswz(arow * 128 + kk * 64 + hi4 * 16));
+      }
+      if (glds_nxt) {
+        conv_stage_a<BM, WAVES>(x, g, abase, k0_nxt, sA(1 - cur), wave,
+                                lane, kfrac, kk, 2);
+        conv_stage_w<BN, WAVES>(wgt + (long long)n0 * kr + k0_nxt, kr,
+                                sB(1 - cur), wave, lane, kk, 2);
+      }
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int i = 0; i < MI; ++i)
+#pragma unroll
+        for (int j = 0; j < NJ; ++j)
+          acc[i][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[i], bfr[j], acc[i][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+#pragma unroll
+  for (int j = 0; j < NJ; ++j) {
+    const int col = n0 + wn * (NJ * 16) + j * 16 + lo;
+    if (!tile_full && col >= n) continue;
+    const float bv = bias ? (float)bias[col] : 0.f;
+#pragma unroll
+    for (int i = 0; i < MI; ++i) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = m0 + wm * (MI * 16) + i * 16 + hi4 * 4 + r;
+        if (!tile_full && row >= mtotal) continue;
+        float v = acc[i][j][r] + bv;
+        if (res) v += (float)res[(long long)row * n + col];
+        out[(long long)row * n + col] = (__bf16)conv_act_f(v, ACT);
+      }
+    }
+  }
+}
+
+template <int ACT>
+void launch_conv(const void* x, const void* w, const void* bias,
+                 const void* res, void* out, const ConvGeom& g,
+                 hipStream_t stream) {
+  const int m = g.b * g.oh * g.ow;
+  const long long tiles_big =
+      (long long)((m + 255) / 256) * ((g.kout + 255) / 256);
+  const bool big = g.kout >= 256 && m >= 4096 && tiles_big >= 150;
+  const int BM = big ? 256 : 128, BN = big ? 256 : 128;
+  const int tiles_m = (m + BM - 1) / BM, tiles_n = (g.kout + BN - 1) / BN;
+  const dim3 grid(tiles_m * tiles_n);
+  const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
+  if (big)
+    hipLaunchKernelGGL((conv2d_nhwc_kernel<ACT, true>), grid, dim3(512),
+                       lds, stream, (const __bf16*)x, (const __bf16*)w,
+                       (const __bf16*)bias, (const __bf16*)res,
+                       (__bf16*)out, g, m, tiles_m, tiles_n);
+  else
+    hipLaunchKernelGGL((conv2d_nhwc_kernel<ACT, false>), grid, dim3(256),
+                       lds, stream, (const __bf16*)x, (const __bf16*)w,
+                       (const __bf16*)bias, (const __bf16*)res,
+                       (__bf16*)out, g, m, tiles_m, tiles_n);
+}
+
+// ------------------------------------------------------------- pad kernel
+// zero-pad H/W of an NHWC tensor: (B, H, W, C) -> (B, H+pt+pb, W+pl+pr, C).
+// Memory-bound; 16-B vectorized over channels (C % 8 == 0).
+__global__ void pad2d_nhwc_kernel(const __bf16* __restrict__ x,
+                                  __bf16* __restrict__ out, int b, int h,
+                                  int w, int c, int pt, int pb, int pl,
+                                  int pr) {
+  const int hp = h + pt + pb, wp = w + pl + pr;
+  const long long total = (long long)b * hp * wp * (c / 8);
+  for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
+       i < total; i += (long long)gridDim.x * blockDim.x) {
+    const int cseg = (int)(i % (c / 8));
+    long long t = i / (c / 8);
+    const int xw = (int)(t % wp);
+    t /= wp;
+    const int xh = (int)(t % hp);
+    const int bi = (int)(t / hp);
+    uint4 v = {0u, 0u, 0u, 0u};
+    const int sy = xh - pt, sx = xw - pl;
+    if (sy >= 0 && sy < h && sx >= 0 && sx < w) {
+      v = *reinterpret_cast<const uint4*>(
+          x + (((long long)bi * h + sy) * w + sx) * c + cseg * 8);
+    }
+    *reinterpret_cast<uint4*>(
+        out + (((long long)bi * hp + xh) * wp + xw) * c + cseg * 8) = v;
+  }
+}
+
+}  // namespace
+
+extern "C" {
+
+// act: 0 none, 1 relu, 2 quick_gelu, 3 gelu_tanh, 4 leaky_relu(0.1)
+void vfa_conv2d_nhwc(const void* x, const void* w, const void* bias,
+                     const void* res, void* out, int b, int hp, int wp,
+                     int cin, int oh, int ow, int kout, int kh, int kw,
+                     int sh, int sw, int act, hipStream_t stream) {
+  ConvGeom g{b, hp, wp, cin, oh, ow, kout, kh, kw, sh, sw, kh * kw * cin};
+  switch (act) {
+    case 0: launch_conv<0>(x, w, bias, res, out, g, stream); break;
+    case 1: launch_conv<1>(x, w, bias, res, out, g, stream); break;
+    case 2: launch_conv<2>(x, w, bias, res, out, g, stream); break;
+    case 3: launch_conv<3>(x, w, bias, res, out, g, stream); break;
+    case 4: launch_conv<4>(x, w, bias, res, out, g, stream); break;
+  }
+}
+
+void vfa_pad2d_nhwc(const void* x, void* out, int b, int h, int w, int c,
+                    int pt, int pb, int pl, int pr, hipStream_t stream) {
+  const long long total = (long long)b * (h + pt + pb) * (w + pl + pr) *
+                          (c / 8);
+  const int threads = 256;
+  const int blocks = (int)min((total + threads - 1) / threads, 2048LL);
+  hipLaunchKernelGGL(pad2d_nhwc_kernel, dim3(blocks), dim3(threads), 0,
+                     stream, (const __bf16*)x, (__bf16*)out, b, h, w, c, pt,
+                     pb, pl, pr);
+}
+
+}  // extern "C"
