@@ -17,11 +17,11 @@ from video_features_amd import ops  # noqa: E402
 
 # (name, B, C, H, W, K, kh, kw, stride, pad)
 SHAPES = [
-    ('rn50 l1 3x3 64>64 56²',    32, 64, 56, 56, 64, 3, 3, 1, (1, 1)),
-    ('rn50 l2 3x3 128>128 28²',  32, 128, 28, 28, 128, 3, 3, 1, (1, 1)),
-    ('rn50 l3 3x3 256>256 14²',  32, 256, 14, 14, 256, 3, 3, 1, (1, 1)),
-    ('rn50 l4 3x3 512>512 7²',   32, 512, 7, 7, 512, 3, 3, 1, (1, 1)),
-    ('rn50 l2 s2 128 56²',       32, 128, 56, 56, 128, 3, 3, 2, (1, 1)),
+    ('rn50 l1 3x3 64>64 56²',   384, 64, 56, 56, 64, 3, 3, 1, (1, 1)),
+    ('rn50 l2 3x3 128>128 28²', 384, 128, 28, 28, 128, 3, 3, 1, (1, 1)),
+    ('rn50 l3 3x3 256>256 14²', 384, 256, 14, 14, 256, 3, 3, 1, (1, 1)),
+    ('rn50 l4 3x3 512>512 7²',  384, 512, 7, 7, 512, 3, 3, 1, (1, 1)),
+    ('rn50 l2 s2 128 56²',      384, 128, 56, 56, 128, 3, 3, 2, (1, 1)),
     ('raft f 3x3 64>64 112²',    16, 64, 112, 112, 64, 3, 3, 1, (1, 1)),
     ('raft f 3x3 96>96 56²',     16, 96, 56, 56, 96, 3, 3, 1, (1, 1)),
     ('raft f 3x3 128>128 28²',   16, 128, 28, 28, 128, 3, 3, 1, (1, 1)),

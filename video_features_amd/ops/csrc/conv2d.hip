@@ -152,22 +152,22 @@ __device__ __forceinline__ void conv_guard_w(
   }
 }
 
-// BIG: 256x256 tile, 8 waves (2Mx4N); else 128x128, 4 waves (2x2).
-template <int ACT, bool BIG>
-__global__ __launch_bounds__(BIG ? 512 : 256)
+// Tile variants: 256x256 (8 waves, 2Mx4N), 128x128 (4 waves, 2x2),
+// 256x64 (4 waves, 4Mx1N — for K_out = 64/192 nets where a 128-wide N
+// tile is half empty).
+template <int ACT, int BM, int BN, int WAVES, int WN>
+__global__ __launch_bounds__(WAVES * 64)
 void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
                         const __bf16* __restrict__ wgt,
                         const __bf16* __restrict__ bias,
                         const __bf16* __restrict__ res,
                         __bf16* __restrict__ out, ConvGeom g, int mtotal,
                         int tiles_m, int tiles_n) {
-  constexpr int BM = BIG ? 256 : 128, BN = BIG ? 256 : 128;
-  constexpr int WAVES = BIG ? 8 : 4;
-  constexpr int WN = BIG ? 4 : 2;
-  constexpr int MI = BIG ? 8 : 4;
-  constexpr int NJ = 4;
+  constexpr int MI = BM / (WAVES / WN) / 16;   // 16-row MFMA tiles / wave
+  constexpr int NJ = BN / WN / 16;             // 16-col MFMA tiles / wave
+  static_assert(NJ == 4, "epilogue assumes 4 column fragments per wave");
   constexpr int TILE_A = BM * BK * 2, TILE_B = BN * BK * 2;
-  constexpr int THREADS = BIG ? 512 : 256;
+  constexpr int THREADS = WAVES * 64;
   constexpr int PER_WAVE = (BM * 128 / 1024) / WAVES;
 
   extern __shared__ __attribute__((aligned(1024))) char smem[];
@@ -287,28 +287,43 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
   }
 }
 
+template <int ACT, int BM, int BN, int WAVES, int WN>
+void launch_cfg(const void* x, const void* w, const void* bias,
+                const void* res, void* out, const ConvGeom& g, int m,
+                hipStream_t stream) {
+  const int tiles_m = (m + BM - 1) / BM, tiles_n = (g.kout + BN - 1) / BN;
+  const dim3 grid(tiles_m * tiles_n);
+  const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
+  hipLaunchKernelGGL((conv2d_nhwc_kernel<ACT, BM, BN, WAVES, WN>), grid,
+                     dim3(WAVES * 64), lds, stream, (const __bf16*)x,
+                     (const __bf16*)w, (const __bf16*)bias,
+                     (const __bf16*)res, (__bf16*)out, g, m, tiles_m,
+                     tiles_n);
+}
+
 template <int ACT>
 void launch_conv(const void* x, const void* w, const void* bias,
                  const void* res, void* out, const ConvGeom& g,
                  hipStream_t stream) {
   const int m = g.b * g.oh * g.ow;
-  const long long tiles_big =
-      (long long)((m + 255) / 256) * ((g.kout + 255) / 256);
-  const bool big = g.kout >= 256 && m >= 4096 && tiles_big >= 150;
-  const int BM = big ? 256 : 128, BN = big ? 256 : 128;
-  const int tiles_m = (m + BM - 1) / BM, tiles_n = (g.kout + BN - 1) / BN;
-  const dim3 grid(tiles_m * tiles_n);
-  const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
-  if (big)
-    hipLaunchKernelGGL((conv2d_nhwc_kernel<ACT, true>), grid, dim3(512),
-                       lds, stream, (const __bf16*)x, (const __bf16*)w,
-                       (const __bf16*)bias, (const __bf16*)res,
-                       (__bf16*)out, g, m, tiles_m, tiles_n);
+  // pick the tile minimizing padded work, preferring the bigger tile when
+  // waste ties and the grid still fills the chip (256 CUs)
+  auto cost = [&](int bm, int bn, long long min_wgs) {
+    const long long tm = (m + bm - 1) / bm, tn = (g.kout + bn - 1) / bn;
+    long long padded = tm * bm * tn * bn;
+    if (tm * tn < min_wgs) padded = padded * 4;   // fill penalty
+    return padded;
+  };
+  const long long c256 = (g.kout >= 256 && m >= 4096)
+                             ? cost(256, 256, 150) : (1LL << 62);
+  const long long c128 = cost(128, 128, 120);
+  const long long c64 = cost(256, 64, 120);
+  if (c256 <= c128 && c256 <= c64)
+    launch_cfg<ACT, 256, 256, 8, 4>(x, w, bias, res, out, g, m, stream);
+  else if (c64 < c128)
+    launch_cfg<ACT, 256, 64, 4, 1>(x, w, bias, res, out, g, m, stream);
   else
-    hipLaunchKernelGGL((conv2d_nhwc_kernel<ACT, false>), grid, dim3(256),
-                       lds, stream, (const __bf16*)x, (const __bf16*)w,
-                       (const __bf16*)bias, (const __bf16*)res,
-                       (__bf16*)out, g, m, tiles_m, tiles_n);
+    launch_cfg<ACT, 128, 128, 4, 2>(x, w, bias, res, out, g, m, stream);
 }
 
 // ------------------------------------------------------------- pad kernel
