@@ -109,8 +109,8 @@ class Unit3D(nn.Module):
                 # bias pass, and the separate ReLU round trip
                 return ops.conv1x1_act(
                     xf, w2, bias, 'relu' if self.activation else 'none')
-            x = F.conv2d(xf, w2, bias,
-                         padding=(self.kernel[1] // 2, self.kernel[2] // 2))
+            x = ops.conv2d_act(xf, w2, bias,
+                               1, (self.kernel[1] // 2, self.kernel[2] // 2))
         else:  # 3x3x3, stride 1
             o = w5.shape[0]
             wcat = cached_cl_weight(
@@ -126,8 +126,11 @@ class Unit3D(nn.Module):
                 bcat = cached_cl_weight(self, 'bcat', bias, mk_bcat)
             else:
                 bcat = None
-            y = F.conv2d(xf, wcat, bcat,
-                         padding=(self.kernel[1] // 2, self.kernel[2] // 2))
+            # merged-tap 3x3 conv (3*O outputs) through the in-tree
+            # implicit-GEMM kernel (1.6-1.9x MIOpen on the I3D shapes,
+            # gpurun_out/bench_conv_r2b.log)
+            y = ops.conv2d_act(xf, wcat, bcat,
+                               1, (self.kernel[1] // 2, self.kernel[2] // 2))
             if self.activation and not isinstance(self.bn, nn.BatchNorm3d):
                 # BN folded: ReLU rides the merge kernel's epilogue
                 return temporal_merge(y, b, kt=3, st=1, p0=1, bias_tap=1,

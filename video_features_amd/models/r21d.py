@@ -16,6 +16,7 @@ import torch
 import torch.nn.functional as F
 from torch import nn
 
+from .. import ops
 from ._flat3d import (flatten_time, temporal_merge, temporal_select,
                       cached_cl_weight)
 
@@ -55,7 +56,7 @@ class Conv2Plus1D(nn.Module):
         ss = self.spatial.stride[1]
         sw = cached_cl_weight(self, 'sw', self.spatial.weight,
                               lambda: self.spatial.weight[:, :, 0])
-        y = F.conv2d(xf, sw, self.spatial.bias, stride=ss, padding=1)
+        y = ops.conv2d_act(xf, sw, self.spatial.bias, ss, 1)
         y = _flat_bn_relu(y, self.bn, True)
         w = self.temporal.weight                   # (O, M, 3, 1, 1)
         o = w.shape[0]
@@ -161,7 +162,7 @@ class R2Plus1D18(nn.Module):
                                 self.stem[4])
         s0w = cached_cl_weight(self, 's0w', sc0.weight,
                                lambda: sc0.weight[:, :, 0])
-        xf = F.conv2d(xf, s0w, sc0.bias, stride=2, padding=3)
+        xf = ops.conv2d_act(xf, s0w, sc0.bias, 2, 3)
         xf = _flat_bn_relu(xf, sbn0, True)
         o = sc1.weight.shape[0]
         wcat = cached_cl_weight(

@@ -81,9 +81,24 @@ class ResidualBlock(nn.Module):
                 nn.Conv2d(in_ch, out_ch, 1, stride), _make_norm(norm, out_ch))
 
     def forward(self, x):
-        y = _norm_act(self.norm1, self.relu, self.conv1(x))
-        y = _norm_act(self.norm2, self.relu, self.conv2(y))
-        identity = x if self.downsample is None else self.downsample(x)
+        # 3x3 convs through the in-tree implicit-GEMM kernel; when the norm
+        # is folded away (cnet's BN) the ReLU rides the conv epilogue.
+        # NOTE the block is relu(x + relu(norm2(conv2(...)))) — the inner
+        # ReLU precedes the residual add (reference extractor.py:47-55),
+        # so the add cannot ride the conv epilogue.
+        n1_id = isinstance(self.norm1, nn.Identity)
+        n2_id = isinstance(self.norm2, nn.Identity)
+        y = ops.conv2d_mod(self.conv1, x, 'relu' if n1_id else 'none')
+        if not n1_id:
+            y = _norm_act(self.norm1, self.relu, y)
+        y = ops.conv2d_mod(self.conv2, y, 'relu' if n2_id else 'none')
+        if not n2_id:
+            y = _norm_act(self.norm2, self.relu, y)
+        if self.downsample is None:
+            identity = x
+        else:
+            identity = self.downsample[1](
+                ops.conv2d_mod(self.downsample[0], x))
         return self.relu(identity + y)
 
 
@@ -145,7 +160,7 @@ class FlowHead(nn.Module):
         self.conv2 = nn.Conv2d(hidden, 2, 3, 1, 1)
 
     def forward(self, x):
-        return self.conv2(F.relu(self.conv1(x)))
+        return self.conv2(ops.conv2d_mod(self.conv1, x, 'relu'))
 
 
 class SepConvGRU(nn.Module):
@@ -182,10 +197,12 @@ class SepConvGRU(nn.Module):
 
     def forward(self, hx: torch.Tensor, rhx: torch.Tensor,
                 nhwc: bool = False) -> None:
-        z = ops.gru_zr(self.convzr1(hx), hx, rhx, nhwc)
-        ops.gru_out(self.convq1(rhx), z, hx, nhwc)
-        z = ops.gru_zr(self.convzr2(hx), hx, rhx, nhwc)
-        ops.gru_out(self.convq2(rhx), z, hx, nhwc)
+        # 1x5 / 5x1 convs through the in-tree implicit-GEMM kernel
+        # (1.8-1.9x MIOpen on the GRU shapes, gpurun_out/bench_conv_r2b.log)
+        z = ops.gru_zr(ops.conv2d_mod(self.convzr1, hx), hx, rhx, nhwc)
+        ops.gru_out(ops.conv2d_mod(self.convq1, rhx), z, hx, nhwc)
+        z = ops.gru_zr(ops.conv2d_mod(self.convzr2, hx), hx, rhx, nhwc)
+        ops.gru_out(ops.conv2d_mod(self.convq2, rhx), z, hx, nhwc)
 
 
 class BasicMotionEncoder(nn.Module):
@@ -201,9 +218,17 @@ class BasicMotionEncoder(nn.Module):
         self.conv = nn.Conv2d(192 + 64, 128 - 2, 3, 1, 1)
 
     def forward(self, flow, corr):
-        c = F.relu(self.convc2(F.relu(self.convc1(corr))))
-        f = F.relu(self.convf2(F.relu(self.convf1(flow))))
-        out = F.relu(self.conv(torch.cat([c, f], dim=1)))
+        c = ops.conv2d_mod(self.convc2,
+                           ops.conv2d_mod(self.convc1, corr, 'relu'),
+                           'relu')
+        f = ops.conv2d_mod(self.convf2,
+                           ops.conv2d_mod(self.convf1, flow, 'relu'),
+                           'relu')
+        cat = torch.cat([c, f], dim=1)
+        if cat.is_cuda and not cat.is_contiguous(
+                memory_format=torch.channels_last):
+            cat = cat.contiguous(memory_format=torch.channels_last)
+        out = ops.conv2d_mod(self.conv, cat, 'relu')
         return torch.cat([out, flow], dim=1)
 
 
@@ -289,6 +314,10 @@ class RAFT(nn.Module):
             rhx[:, ctx_end:] = motion
             ub.gru(hx, rhx, self.nhwc)
             net = hx[:, :self.hdim]
+            if self.nhwc:
+                # the channel slice of the CL buffer is strided; one copy
+                # lets flow_head/mask take the in-tree conv kernel
+                net = net.contiguous(memory_format=torch.channels_last)
             delta = ub.flow_head(net)
             coords1 = (coords1 + delta.float()).contiguous()
             if it == iters - 1 or not test_mode:

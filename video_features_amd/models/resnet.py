@@ -37,6 +37,19 @@ class BasicBlock(nn.Module):
                 nn.BatchNorm2d(planes))
 
     def forward(self, x):
+        if (isinstance(self.bn1, nn.Identity) and x.is_cuda
+                and x.dtype == torch.bfloat16
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and ops.hip_available()):
+            # both 3x3 convs through the in-tree implicit-GEMM kernel;
+            # conv2 fuses the residual add + ReLU into its epilogue
+            idt = x if self.downsample is None else self.downsample(x)
+            if not idt.is_contiguous(memory_format=torch.channels_last):
+                idt = idt.contiguous(memory_format=torch.channels_last)
+            y = ops.conv2d_act(x, self.conv1.weight, self.conv1.bias,
+                               self.conv1.stride[0], 1, 'relu')
+            return ops.conv2d_act(y, self.conv2.weight, self.conv2.bias,
+                                  1, 1, 'relu', res=idt)
         identity = x if self.downsample is None else self.downsample(x)
         out = self.relu(self.bn1(self.conv1(x)))
         out = self.bn2(self.conv2(out))
@@ -84,9 +97,11 @@ class Bottleneck(nn.Module):
             else:
                 y = F.relu(F.conv2d(x, self.conv1.weight, self.conv1.bias),
                            inplace=True)
-            y = F.relu(F.conv2d(y, self.conv2.weight, self.conv2.bias,
-                                stride=self.conv2.stride, padding=1),
-                       inplace=True)
+            # 3x3 conv + bias + ReLU as ONE in-tree implicit-GEMM MFMA
+            # kernel (conv2d.hip — 1.4-1.5x MIOpen on these shapes,
+            # gpurun_out/bench_conv_r2b.log)
+            y = ops.conv2d_act(y, self.conv2.weight, self.conv2.bias,
+                               self.conv2.stride[0], 1, 'relu')
             if not idt.is_contiguous(memory_format=torch.channels_last):
                 idt = idt.contiguous(memory_format=torch.channels_last)
             return ops.conv1x1_act(y, self.conv3.weight, self.conv3.bias,

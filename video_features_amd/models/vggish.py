@@ -119,7 +119,25 @@ class VGGishNet(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         """(N, 96, 64) log-mel examples → (N, 128) embeddings."""
-        x = self.features(x[:, None])                # (N, 512, 6, 4)
+        x = x[:, None]
+        if (x.is_cuda and x.dtype == torch.bfloat16
+            and __import__('video_features_amd.ops', fromlist=['o'])
+                .hip_available()):
+            from .. import ops
+            # conv+ReLU pairs through the in-tree implicit-GEMM kernel
+            # (first conv has C=1 → eager fallback inside conv2d_mod)
+            x = x.contiguous(memory_format=torch.channels_last)
+            mods = list(self.features)
+            i = 0
+            while i < len(mods):
+                if isinstance(mods[i], nn.Conv2d):
+                    x = ops.conv2d_mod(mods[i], x, 'relu')
+                    i += 2                       # skip the fused ReLU
+                else:
+                    x = mods[i](x)
+                    i += 1
+        else:
+            x = self.features(x)                 # (N, 512, 6, 4)
         # TF-order flatten (reference vggish.py:22-29)
         x = x.permute(0, 2, 3, 1).contiguous().flatten(1)
         return self.embeddings(x)

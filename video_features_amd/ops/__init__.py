@@ -454,6 +454,40 @@ def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
     return y
 
 
+def conv2d_mod(conv: torch.nn.Module, x: torch.Tensor, act: str = 'none',
+               res: Optional[torch.Tensor] = None) -> torch.Tensor:
+    """Route an ``nn.Conv2d`` module call through the in-tree kernels:
+    1x1 → fused MFMA GEMM (conv1x1_act), kxk → implicit-GEMM conv
+    (conv2d_act); eager composition otherwise (CPU, C % 8 != 0, dilation,
+    groups, tiny K_out)."""
+    w = conv.weight
+    kh, kw = w.shape[2], w.shape[3]
+    eligible = (x.is_cuda and x.dtype == torch.bfloat16
+                and w.dtype == torch.bfloat16
+                and x.shape[1] % 8 == 0 and w.shape[0] >= 16
+                and conv.dilation == (1, 1) and conv.groups == 1
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and hip_available() and not _env_flag('VFA_NO_CONV'))
+    if eligible and kh == 1 and kw == 1 and conv.stride == (1, 1) \
+            and w.shape[0] % 8 == 0:
+        return conv1x1_act(x, w, conv.bias, act, res)
+    if eligible:
+        return conv2d_act(x, w, conv.bias, conv.stride, conv.padding, act,
+                          res)
+    y = conv(x)
+    if res is not None:
+        y = y + res
+    if act == 'relu':
+        y = torch.nn.functional.relu(y)
+    elif act == 'leaky_relu':
+        y = torch.nn.functional.leaky_relu(y, 0.1)
+    elif act == 'quick_gelu':
+        y = y * torch.sigmoid(1.702 * y)
+    elif act == 'gelu':
+        y = torch.nn.functional.gelu(y, approximate='tanh')
+    return y
+
+
 def grid_sample_bilinear(x: torch.Tensor, coords: torch.Tensor) -> torch.Tensor:
     """RAFT-style bilinear lookup: ``coords`` (B, Ho, Wo, 2) in *pixel* units,
     zero padding outside (reference models/raft/raft_src/utils/utils.py:57-71)."""
