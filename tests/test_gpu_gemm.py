@@ -51,6 +51,43 @@ def test_linear_act(dev, m, n, k, act):
     assert (err.mean() / ref.abs().mean().clamp_min(1e-6)).item() < 5e-3
 
 
+@pytest.mark.parametrize('m,n,k', [
+    (131072, 64, 64),     # resnet l1 conv1 regime (thin-K streaming path)
+    (131072, 256, 64),    # l1 conv3
+    (131072, 128, 256),   # l2 conv1
+    (70000, 512, 128),    # l2 conv3 (ragged M)
+    (65552, 48, 96),      # ragged N + M just over the threshold
+])
+@pytest.mark.parametrize('act', ['none', 'relu'])
+def test_linear_thin_streaming(dev, m, n, k, act):
+    """The M-huge / K-shallow streaming kernel (1x1-conv-as-GEMM shapes)."""
+    ops = _hip_loaded()
+    torch.manual_seed(0)
+    x = (torch.randn(m, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    b = torch.randn(n, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, b, act)
+    ref = _ref(x, w, b, act)
+    err = (out.float() - ref).abs()
+    rel = err.max().item() / max(ref.abs().max().item(), 1e-6)
+    assert rel < 3e-2, rel
+
+
+def test_linear_thin_residual(dev):
+    """bottleneck conv3 epilogue: bias + residual + relu on the thin path."""
+    ops = _hip_loaded()
+    torch.manual_seed(3)
+    m, n, k = 131072, 256, 64
+    x = (torch.randn(m, k, device=dev) / 2).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) / 2).to(torch.bfloat16)
+    b = torch.randn(n, device=dev).to(torch.bfloat16)
+    r = torch.randn(m, n, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, b, 'relu', r)
+    ref = (_ref(x, w, b, 'none') + r.float()).relu()
+    rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert rel < 3e-2, rel
+
+
 def test_linear_act_no_bias(dev):
     ops = _hip_loaded()
     torch.manual_seed(1)

@@ -27,6 +27,11 @@ SHAPES = [
     (8192, 8192, 8192, 'none', 'ladder 8192^3'),
     (19200, 3072, 768, 'quick_gelu', 'CLIP fc1 (fb384)'),
     (19200, 2304, 768, 'none', 'CLIP qkv (fb384)'),
+    (1204224, 64, 64, 'relu', 'rn50 l1 conv1 (thin)'),
+    (1204224, 256, 64, 'relu', 'rn50 l1 conv3 (thin)'),
+    (1204224, 128, 256, 'relu', 'rn50 l2 conv1 (thin)'),
+    (301056, 512, 128, 'relu', 'rn50 l2 conv3 (thin)'),
+    (301056, 256, 512, 'relu', 'rn50 l3 conv1 (not thin)'),
 ]
 
 

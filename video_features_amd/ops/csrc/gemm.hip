@@ -234,6 +234,112 @@ void linear_act_kernel(const __bf16* __restrict__ a,
   }
 }
 
+// ------------------------------------------------------- thin-K streaming
+// M-huge / K-shallow GEMMs (the ResNet/I3D 1x1 convs as GEMM: M = B*H*W up
+// to ~1.2M rows, K = 64..256) are HBM-streaming problems: the tiled
+// double-buffered kernel above degenerates to a 1-iteration K-loop and
+// runs ~2x off the bandwidth floor (profiles/, round 2).  This kernel
+// keeps W resident in LDS (loaded once per block), streams A directly
+// global -> MFMA fragments (A is read ONCE for all N columns), grid-strides
+// over M, and has no per-K barriers.  vgpr budget: A frags 2*K/32*4 +
+// B frags K/32*4 + acc 8 -> ~130 at K=256 (2 blocks/CU).
+template <int ACT, int KMAX>
+__global__ __launch_bounds__(256, 2)
+void linear_thin_kernel(const __bf16* __restrict__ a,
+                        const __bf16* __restrict__ w,
+                        const __bf16* __restrict__ bias,
+                        const __bf16* __restrict__ res,
+                        __bf16* __restrict__ c, int m, int n, int k,
+                        int nch) {
+  // grid: (m_blocks, n_outer); block 256 = 4 waves, each wave 32 rows
+  constexpr int KF_MAX = KMAX / 32;            // K fragments (<= 8)
+  const int kf = k / 32;
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  const int lo = lane & 15, hi4 = lane >> 4;
+
+  // W chunk for this n_outer: rows [n0, n0+nch) of (N, K), padded LDS rows
+  // of k+8 elems to break the ds_read bank pattern
+  extern __shared__ __attribute__((aligned(16))) char smem_t[];
+  __bf16* wl = reinterpret_cast<__bf16*>(smem_t);
+  const int ldw = k + 8;
+  const int n0 = blockIdx.y * nch;
+  const int ncols = min(nch, n - n0);
+  for (int t = threadIdx.x; t < ncols * (k / 8); t += 256) {
+    const int row = t / (k / 8), seg = t % (k / 8);
+    *reinterpret_cast<uint4*>(wl + row * ldw + seg * 8) =
+        *reinterpret_cast<const uint4*>(w + (long long)(n0 + row) * k +
+                                        seg * 8);
+  }
+  __syncthreads();
+
+  const long long mstep = (long long)gridDim.x * 128;
+  for (long long m0 = (long long)blockIdx.x * 128; m0 < m; m0 += mstep) {
+    const long long r0 = m0 + wave * 32;
+    // A fragments for this wave's 32 rows (2 x 16), K resident in regs
+    bf16x8 afr[2][KF_MAX];
+#pragma unroll
+    for (int mi = 0; mi < 2; ++mi) {
+      long long row = r0 + mi * 16 + lo;
+      if (row >= m) row = m - 1;               // clamped load, masked store
+      const __bf16* ap = a + row * k + hi4 * 8;
+      for (int kk = 0; kk < kf; ++kk)
+        afr[mi][kk] = *reinterpret_cast<const bf16x8*>(ap + kk * 32);
+    }
+    for (int j = 0; j < ncols; j += 16) {
+      f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
+      const __bf16* wp = wl + (j + lo) * ldw + hi4 * 8;
+      for (int kk = 0; kk < kf; ++kk) {
+        const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(wp + kk * 32);
+#pragma unroll
+        for (int mi = 0; mi < 2; ++mi)
+          acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mi][kk], bfr, acc[mi], 0, 0, 0);
+      }
+      const int col = n0 + j + lo;
+      const bool colok = col < n && j + lo < ncols;
+      const float bv = (bias && colok) ? (float)bias[col] : 0.f;
+#pragma unroll
+      for (int mi = 0; mi < 2; ++mi) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const long long row = r0 + mi * 16 + hi4 * 4 + r;
+          if (row >= m || !colok) continue;
+          float v = acc[mi][r] + bv;
+          if (res) v += (float)res[row * n + col];
+          c[row * n + col] = (__bf16)act_f(v, ACT);
+        }
+      }
+    }
+  }
+}
+
+template <int ACT>
+bool launch_thin(const void* a, const void* w, const void* bias,
+                 const void* res, void* c, int m, int n, int k,
+                 hipStream_t stream) {
+  if (k > 256 || k % 32 != 0 || m < 65536) return false;
+  // W chunk bounded by 64 KiB LDS (2 blocks/CU)
+  const int nch_cap = 32768 / (k + 8) / 16 * 16;
+  const int nch = min(n + 15 & ~15, nch_cap);
+  if (nch < 16) return false;
+  const int n_outer = (n + nch - 1) / nch;
+  // enough M-blocks to fill the chip; grid-stride handles the rest
+  const int m_blocks = (int)min(((long long)m + 127) / 128, 4096LL);
+  const size_t lds = (size_t)nch * (k + 8) * 2;
+  const dim3 grid(m_blocks, n_outer);
+  if (k <= 128)
+    hipLaunchKernelGGL((linear_thin_kernel<ACT, 128>), grid, dim3(256),
+                       lds, stream, (const __bf16*)a, (const __bf16*)w,
+                       (const __bf16*)bias, (const __bf16*)res, (__bf16*)c,
+                       m, n, k, nch);
+  else
+    hipLaunchKernelGGL((linear_thin_kernel<ACT, 256>), grid, dim3(256),
+                       lds, stream, (const __bf16*)a, (const __bf16*)w,
+                       (const __bf16*)bias, (const __bf16*)res, (__bf16*)c,
+                       m, n, k, nch);
+  return true;
+}
+
 template <int ACT, bool BIG>
 void launch_tile(const void* a, const void* w, const void* bias,
                  const void* res, void* c, int m, int n, int k,
@@ -261,9 +367,12 @@ template <int ACT>
 void launch_linear(const void* a, const void* w, const void* bias,
                    const void* res, void* c, int m, int n, int k,
                    hipStream_t stream) {
-  // BIG tiles when M tiles evenly (a half-empty 256-row tail tile and the
-  // block-round quantization cost more than the smaller tile's overhead —
-  // measured 339 vs 528 TF at M=9600) and the grid still fills the chip
+  // M-huge / K-shallow -> the streaming kernel (reads A once, no K-loop
+  // barriers); else BIG tiles when M tiles evenly (a half-empty 256-row
+  // tail tile and the block-round quantization cost more than the smaller
+  // tile's overhead — measured 339 vs 528 TF at M=9600) and the grid
+  // still fills the chip
+  if (launch_thin<ACT>(a, w, bias, res, c, m, n, k, stream)) return;
   const long long tiles = (long long)((m + 255) / 256) * ((n + 255) / 256);
   if (n >= 256 && m % 256 == 0 && tiles >= 150)
     launch_tile<ACT, true>(a, w, bias, res, c, m, n, k, stream);
