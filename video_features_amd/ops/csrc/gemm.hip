@@ -243,31 +243,34 @@ void linear_act_kernel(const __bf16* __restrict__ a,
 // global -> MFMA fragments (A is read ONCE for all N columns), grid-strides
 // over M, and has no per-K barriers.  vgpr budget: A frags 2*K/32*4 +
 // B frags K/32*4 + acc 8 -> ~130 at K=256 (2 blocks/CU).
-template <int ACT, int KMAX>
+template <int ACT, int KF>
 __global__ __launch_bounds__(256, 2)
 void linear_thin_kernel(const __bf16* __restrict__ a,
                         const __bf16* __restrict__ w,
                         const __bf16* __restrict__ bias,
                         const __bf16* __restrict__ res,
-                        __bf16* __restrict__ c, int m, int n, int k,
-                        int nch) {
-  // grid: (m_blocks, n_outer); block 256 = 4 waves, each wave 32 rows
-  constexpr int KF_MAX = KMAX / 32;            // K fragments (<= 8)
-  const int kf = k / 32;
+                        __bf16* __restrict__ c, int m, int n, int nch) {
+  // grid: (m_blocks, n_outer); block 256 = 4 waves, each wave 32 rows.
+  // KF = K/32 is COMPILE-TIME: runtime-indexed ext_vector arrays go to
+  // scratch (cdna_hip_programming.md §5.4 rule 20 — the first version of
+  // this kernel hit exactly that: 4 TF/s from scratch traffic)
+  constexpr int K = KF * 32;
   const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   const int lo = lane & 15, hi4 = lane >> 4;
 
-  // W chunk for this n_outer: rows [n0, n0+nch) of (N, K), padded LDS rows
-  // of k+8 elems to break the ds_read bank pattern
+  // LDS layout: W chunk (rows of K+8 elems, breaks the ds_read bank
+  // pattern) + per-wave 16x64 bf16 epilogue bounce tiles (row pitch 72)
   extern __shared__ __attribute__((aligned(16))) char smem_t[];
   __bf16* wl = reinterpret_cast<__bf16*>(smem_t);
-  const int ldw = k + 8;
+  constexpr int LDW = K + 8;
   const int n0 = blockIdx.y * nch;
   const int ncols = min(nch, n - n0);
-  for (int t = threadIdx.x; t < ncols * (k / 8); t += 256) {
-    const int row = t / (k / 8), seg = t % (k / 8);
-    *reinterpret_cast<uint4*>(wl + row * ldw + seg * 8) =
-        *reinterpret_cast<const uint4*>(w + (long long)(n0 + row) * k +
+  __bf16* et = reinterpret_cast<__bf16*>(smem_t) + (long long)nch * LDW +
+               wave * 16 * 72;
+  for (int t = threadIdx.x; t < ncols * (K / 8); t += 256) {
+    const int row = t / (K / 8), seg = t % (K / 8);
+    *reinterpret_cast<uint4*>(wl + row * LDW + seg * 8) =
+        *reinterpret_cast<const uint4*>(w + (long long)(n0 + row) * K +
                                         seg * 8);
   }
   __syncthreads();
@@ -276,67 +279,129 @@ void linear_thin_kernel(const __bf16* __restrict__ a,
   for (long long m0 = (long long)blockIdx.x * 128; m0 < m; m0 += mstep) {
     const long long r0 = m0 + wave * 32;
     // A fragments for this wave's 32 rows (2 x 16), K resident in regs
-    bf16x8 afr[2][KF_MAX];
+    bf16x8 afr[2][KF];
 #pragma unroll
     for (int mi = 0; mi < 2; ++mi) {
       long long row = r0 + mi * 16 + lo;
       if (row >= m) row = m - 1;               // clamped load, masked store
-      const __bf16* ap = a + row * k + hi4 * 8;
-      for (int kk = 0; kk < kf; ++kk)
+      const __bf16* ap = a + row * K + hi4 * 8;
+#pragma unroll
+      for (int kk = 0; kk < KF; ++kk)
         afr[mi][kk] = *reinterpret_cast<const bf16x8*>(ap + kk * 32);
     }
-    for (int j = 0; j < ncols; j += 16) {
-      f32x4 acc[2] = {{0.f, 0.f, 0.f, 0.f}, {0.f, 0.f, 0.f, 0.f}};
-      const __bf16* wp = wl + (j + lo) * ldw + hi4 * 8;
-      for (int kk = 0; kk < kf; ++kk) {
-        const bf16x8 bfr = *reinterpret_cast<const bf16x8*>(wp + kk * 32);
+    // 64 columns per outer step: the epilogue then sees 128-B contiguous
+    // output rows (the naive fragment-layout epilogue's 2-B scalar
+    // res-gathers/stores ran 3x slower than the MFMA+stream itself)
+    for (int j0 = 0; j0 < ncols; j0 += 64) {
+      f32x4 acc[2][4];
 #pragma unroll
-        for (int mi = 0; mi < 2; ++mi)
-          acc[mi] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[mi][kk], bfr, acc[mi], 0, 0, 0);
+      for (int mi = 0; mi < 2; ++mi)
+#pragma unroll
+        for (int jj = 0; jj < 4; ++jj) acc[mi][jj] = f32x4{0, 0, 0, 0};
+      const int njj = min(4, (ncols - j0 + 15) / 16);
+      // jj stays COMPILE-TIME (runtime-indexed acc -> scratch, rule 20);
+      // the tail guard is a wave-uniform predicate around each step
+#pragma unroll
+      for (int jj = 0; jj < 4; ++jj) {
+        if (jj < njj) {
+          const __bf16* wp = wl + (j0 + jj * 16 + lo) * LDW + hi4 * 8;
+#pragma unroll
+          for (int kk = 0; kk < KF; ++kk) {
+            const bf16x8 bfr =
+                *reinterpret_cast<const bf16x8*>(wp + kk * 32);
+            acc[0][jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[0][kk], bfr, acc[0][jj], 0, 0, 0);
+            acc[1][jj] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[1][kk], bfr, acc[1][jj], 0, 0, 0);
+          }
+        }
       }
-      const int col = n0 + j + lo;
-      const bool colok = col < n && j + lo < ncols;
-      const float bv = (bias && colok) ? (float)bias[col] : 0.f;
+      const int jcols = min(64, ncols - j0);   // columns this step
 #pragma unroll
       for (int mi = 0; mi < 2; ++mi) {
+        // fragment -> LDS tile (2-B writes; wave-local, no barrier)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const long long row = r0 + mi * 16 + hi4 * 4 + r;
-          if (row >= m || !colok) continue;
-          float v = acc[mi][r] + bv;
-          if (res) v += (float)res[row * n + col];
-          c[row * n + col] = (__bf16)act_f(v, ACT);
+        for (int jj = 0; jj < 4; ++jj)
+#pragma unroll
+          for (int r = 0; r < 4; ++r)
+            et[(hi4 * 4 + r) * 72 + jj * 16 + lo] =
+                (__bf16)acc[mi][jj][r];
+        __builtin_amdgcn_s_waitcnt(/*lgkmcnt(0)*/ 0xc07f);
+        // vectorized out: 2 passes x 8 rows; lane -> (row, 8-col seg),
+        // 16-B res loads + stores on 128-B-contiguous rows
+#pragma unroll
+        for (int p = 0; p < 2; ++p) {
+          const int er = p * 8 + (lane >> 3);
+          const int ec = (lane & 7) * 8;
+          const long long row = r0 + mi * 16 + er;
+          const int col = n0 + j0 + ec;
+          if (row >= m || ec >= jcols) continue;
+          bf16x8 v8 = *reinterpret_cast<const bf16x8*>(et + er * 72 + ec);
+          if (ec + 8 <= jcols && col + 8 <= n) {
+            bf16x8 o8;
+            bf16x8 b8{}, rr8{};
+            if (bias)
+              b8 = *reinterpret_cast<const bf16x8*>(bias + col);
+            if (res)
+              rr8 = *reinterpret_cast<const bf16x8*>(res + row * n + col);
+#pragma unroll
+            for (int e = 0; e < 8; ++e) {
+              float v = (float)v8[e];
+              if (bias) v += (float)b8[e];
+              if (res) v += (float)rr8[e];
+              o8[e] = (__bf16)act_f(v, ACT);
+            }
+            *reinterpret_cast<bf16x8*>(c + row * n + col) = o8;
+          } else {
+            for (int e = 0; e < 8 && ec + e < jcols && col + e < n; ++e) {
+              float v = (float)v8[e];
+              if (bias) v += (float)bias[col + e];
+              if (res) v += (float)res[row * n + col + e];
+              c[row * n + col + e] = (__bf16)act_f(v, ACT);
+            }
+          }
         }
+        if (mi == 0) __builtin_amdgcn_s_waitcnt(0xc07f);  // tile reuse
       }
     }
   }
+}
+
+template <int ACT, int KF>
+void launch_thin_kf(const void* a, const void* w, const void* bias,
+                    const void* res, void* c, int m, int n, int nch,
+                    int m_blocks, size_t lds, hipStream_t stream) {
+  hipLaunchKernelGGL((linear_thin_kernel<ACT, KF>),
+                     dim3(m_blocks, (n + nch - 1) / nch), dim3(256), lds,
+                     stream, (const __bf16*)a, (const __bf16*)w,
+                     (const __bf16*)bias, (const __bf16*)res, (__bf16*)c,
+                     m, n, nch);
 }
 
 template <int ACT>
 bool launch_thin(const void* a, const void* w, const void* bias,
                  const void* res, void* c, int m, int n, int k,
                  hipStream_t stream) {
-  if (k > 256 || k % 32 != 0 || m < 65536) return false;
-  // W chunk bounded by 64 KiB LDS (2 blocks/CU)
-  const int nch_cap = 32768 / (k + 8) / 16 * 16;
-  const int nch = min(n + 15 & ~15, nch_cap);
+  // n % 8: the vectorized epilogue's 16-B res/out accesses need 8-elem
+  // row alignment
+  if (k > 256 || k % 32 != 0 || m < 65536 || n % 8 != 0) return false;
+  // W chunk + 9 KiB epilogue tiles bounded by 80 KiB LDS (2 blocks/CU)
+  const int nch_cap = 36352 / (k + 8) / 16 * 16;
+  const int nch = min((n + 15) & ~15, nch_cap);
   if (nch < 16) return false;
-  const int n_outer = (n + nch - 1) / nch;
   // enough M-blocks to fill the chip; grid-stride handles the rest
   const int m_blocks = (int)min(((long long)m + 127) / 128, 4096LL);
-  const size_t lds = (size_t)nch * (k + 8) * 2;
-  const dim3 grid(m_blocks, n_outer);
-  if (k <= 128)
-    hipLaunchKernelGGL((linear_thin_kernel<ACT, 128>), grid, dim3(256),
-                       lds, stream, (const __bf16*)a, (const __bf16*)w,
-                       (const __bf16*)bias, (const __bf16*)res, (__bf16*)c,
-                       m, n, k, nch);
-  else
-    hipLaunchKernelGGL((linear_thin_kernel<ACT, 256>), grid, dim3(256),
-                       lds, stream, (const __bf16*)a, (const __bf16*)w,
-                       (const __bf16*)bias, (const __bf16*)res, (__bf16*)c,
-                       m, n, k, nch);
+  const size_t lds = (size_t)nch * (k + 8) * 2 + 4 * 16 * 72 * 2;
+  switch (k / 32) {
+    case 2: launch_thin_kf<ACT, 2>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    case 3: launch_thin_kf<ACT, 3>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    case 4: launch_thin_kf<ACT, 4>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    case 5: launch_thin_kf<ACT, 5>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    case 6: launch_thin_kf<ACT, 6>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    case 7: launch_thin_kf<ACT, 7>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    case 8: launch_thin_kf<ACT, 8>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
+    default: return false;                     // K=32: not worth MFMA
+  }
   return true;
 }
 
