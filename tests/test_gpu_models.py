@@ -53,6 +53,24 @@ def test_i3d_gpu_vs_cpu(dev):
     assert _cos(out, ref) > 0.999
 
 
+@pytest.mark.parametrize('modality,t', [('rgb', 16), ('flow', 15)])
+def test_i3d_gpu_bf16_flat_stem(dev, modality, t):
+    """bf16 GPU path: the 7x7x7 stem runs as the channel-padded 7-tap
+    merged conv2d + strided temporal merge (no conv3d at all); odd T
+    exercises the asymmetric TF-SAME temporal pad."""
+    from video_features_amd.models.i3d import I3D
+    torch.manual_seed(0)
+    c = 3 if modality == 'rgb' else 2
+    m = I3D(modality=modality).eval()
+    x = torch.randn(1, c, t, 128, 128)
+    with torch.no_grad():
+        ref = m.forward_features(x)
+        out = m.to(dev).to(torch.bfloat16).forward_features(
+            x.to(dev).to(torch.bfloat16)).float().cpu()
+    assert out.shape == (1, 1024)
+    assert _cos(out, ref) > 0.99
+
+
 def test_resnet50_gpu_vs_cpu(dev):
     from video_features_amd.models.resnet import build_resnet
     torch.manual_seed(0)

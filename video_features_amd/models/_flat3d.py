@@ -62,7 +62,7 @@ def cl_empty(bt: int, c: int, h: int, w: int, like: torch.Tensor):
 
 def temporal_merge(y: torch.Tensor, b: int, kt: int, st: int = 1,
                    p0: int = 1, bias_tap: int = None,
-                   relu: bool = False) -> torch.Tensor:
+                   relu: bool = False, p1: int = None) -> torch.Tensor:
     """y (B*T, kt*O, H, W): temporal-tap conv outputs stacked along channels
     → (B*T', O, H, W) with out[to] = Σ_dt y_dt[to*st - p0 + dt] (zero
     temporal padding).  ``bias_tap``: the tap that already carries the conv
@@ -71,10 +71,12 @@ def temporal_merge(y: torch.Tensor, b: int, kt: int, st: int = 1,
     bt, ckt, h, w = y.shape
     o = ckt // kt
     t = bt // b
-    to = (t + 2 * p0 - kt) // st + 1
+    if p1 is None:
+        p1 = p0
+    to = (t + p0 + p1 - kt) // st + 1
     if bias_tap is None:
         bias_tap = kt // 2
-    fused = ops.temporal_merge_fused(y, b, kt, st, p0, relu)
+    fused = ops.temporal_merge_fused(y, b, kt, st, p0, relu, p1)
     if fused is not None:
         return fused
     out = cl_empty(b * to, o, h, w, y)

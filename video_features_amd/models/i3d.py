@@ -227,8 +227,52 @@ class I3D(nn.Module):
 
     def _backbone(self, x: torch.Tensor) -> torch.Tensor:
         b = x.shape[0]
-        x = self.conv3d_1a_7x7(x)              # stem stays a real conv3d
-        xf = flatten_time(x)                   # (B*T, C, H, W) channels_last
+        stem = self.conv3d_1a_7x7
+        if (x.is_cuda and x.dtype == torch.bfloat16 and ops.hip_available()
+                and not ops._env_flag('VFA_NO_CONV')):
+            # stem 7x7x7/2 decomposed like the 3x3x3 units: flatten time,
+            # ONE merged 7-tap conv2d (channel-padded C=3/2 -> 8, through
+            # the in-tree implicit-GEMM kernel) + strided temporal
+            # shift-add — the last real conv3d (MIOpen/CK) is gone
+            t, hh, ww = x.shape[2:]
+            xf = flatten_time(x)
+            w5 = stem.conv.weight              # (64, C, 7, 7, 7)
+            cin = w5.shape[1]
+            c8 = (cin + 7) // 8 * 8
+            wcat = cached_cl_weight(
+                stem, 'stem_wcat', w5,
+                lambda: F.pad(
+                    w5.permute(2, 0, 1, 3, 4).reshape(7 * 64, cin, 7, 7),
+                    (0, 0, 0, 0, 0, c8 - cin)))
+            bias = stem.conv.bias
+            if bias is not None:
+                def mk_b():
+                    z = torch.zeros(7 * 64, device=bias.device,
+                                    dtype=bias.dtype)
+                    # tap 3 is valid at every output position for the
+                    # TF-SAME (front 2/3) stride-2 temporal pad
+                    z[3 * 64:4 * 64] = bias
+                    return z
+                bcat = cached_cl_weight(stem, 'stem_bcat', bias, mk_b)
+            else:
+                bcat = None
+            pt = _same_pad_1d(t, 7, 2)
+            ph = _same_pad_1d(hh, 7, 2)
+            pw = _same_pad_1d(ww, 7, 2)
+            y = ops.conv2d_act(xf, wcat, bcat, 2,
+                               (ph[0], ph[1], pw[0], pw[1]))
+            bn_folded = not isinstance(stem.bn, nn.BatchNorm3d)
+            xf = temporal_merge(y, b, kt=7, st=2, p0=pt[0], bias_tap=3,
+                                relu=bn_folded, p1=pt[1])
+            if not bn_folded:
+                bn = stem.bn
+                xf = F.relu(
+                    F.batch_norm(xf, bn.running_mean, bn.running_var,
+                                 bn.weight, bn.bias, False, bn.momentum,
+                                 bn.eps), inplace=True)
+        else:
+            x = self.conv3d_1a_7x7(x)          # CPU path: real conv3d
+            xf = flatten_time(x)               # (B*T, C, H, W) CL
         xf = self.maxPool3d_2a_3x3.forward_flat(xf, b)
         xf = self.conv3d_2b_1x1.forward_flat(xf, b)
         xf = self.conv3d_2c_3x3.forward_flat(xf, b)

@@ -119,9 +119,11 @@ class BasicEncoder(nn.Module):
         self.conv2 = nn.Conv2d(128, output_dim, 1)
 
     def forward(self, x):
-        x = _norm_act(self.norm1, self.relu1, self.conv1(x))
+        # stem 7x7 (C=3) through the in-tree kernel via channel padding
+        x = _norm_act(self.norm1, self.relu1,
+                      ops.conv2d_mod(self.conv1, x))
         x = self.layer3(self.layer2(self.layer1(x)))
-        return self.conv2(x)
+        return ops.conv2d_mod(self.conv2, x)
 
 
 # ------------------------------------------------------------ correlation

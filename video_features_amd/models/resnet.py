@@ -136,7 +136,13 @@ class ResNet(nn.Module):
         return nn.Sequential(*blocks)
 
     def forward_features(self, x: torch.Tensor) -> torch.Tensor:
-        x = self.maxpool(self.relu(self.bn1(self.conv1(x))))
+        # stem 7x7 (C=3) through the in-tree kernel via channel padding;
+        # ReLU rides the conv epilogue once BN is folded
+        bn1_id = isinstance(self.bn1, nn.Identity)
+        x = ops.conv2d_mod(self.conv1, x, 'relu' if bn1_id else 'none')
+        if not bn1_id:
+            x = self.relu(self.bn1(x))
+        x = self.maxpool(x)
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         return self.avgpool(x).flatten(1)
 

@@ -390,14 +390,16 @@ def linear_act(x: torch.Tensor, weight: torch.Tensor,
 
 
 def temporal_merge_fused(y: torch.Tensor, b: int, kt: int, st: int,
-                         p0: int, relu: bool = False) -> Optional[torch.Tensor]:
+                         p0: int, relu: bool = False,
+                         p1: Optional[int] = None) -> Optional[torch.Tensor]:
     """One-kernel temporal tap merge for the flattened-time conv3d
     decomposition (see models/_flat3d.py), with optional fused ReLU.
     Returns None when the HIP path does not apply (caller falls back to
     the strided-add composition)."""
     if (_use_hip(y)
             and y.is_contiguous(memory_format=torch.channels_last)):
-        return _ext.temporal_merge(y, b, kt, st, p0, relu)
+        return _ext.temporal_merge(y, b, kt, st, p0,
+                                   p0 if p1 is None else p1, relu)
     return None
 
 
@@ -429,18 +431,31 @@ def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
     F.conv2d composition.
     """
     sh, sw = (stride, stride) if isinstance(stride, int) else stride
-    ph, pw = (padding, padding) if isinstance(padding, int) else padding
+    if isinstance(padding, int):
+        pt = pb = pl = pr = padding
+    elif len(padding) == 2:
+        pt = pb = padding[0]
+        pl = pr = padding[1]
+    else:                      # 4-way (TF-SAME even-input stems): t,b,l,r
+        pt, pb, pl, pr = padding
+    # weight.shape[1] may be x's channel count zero-padded up to 8 (the
+    # caller pads the weight; the kernel's pad pass widens the input)
     if (_use_hip(x) and x.dtype == torch.bfloat16
             and weight.dtype == torch.bfloat16
-            and x.shape[1] % 8 == 0
+            and weight.shape[1] == (x.shape[1] + 7) // 8 * 8
             and x.is_contiguous(memory_format=torch.channels_last)
             and not _env_flag('VFA_NO_CONV')):
         w_cl = weight.contiguous(memory_format=torch.channels_last)
         r = res.contiguous(memory_format=torch.channels_last) \
             if res is not None else None
-        return _ext.conv2d_nhwc(x, w_cl, bias, r, sh, sw, ph, pw,
+        return _ext.conv2d_nhwc(x, w_cl, bias, r, sh, sw, pt, pb, pl, pr,
                                 _ACT_IDS[act])
-    y = torch.nn.functional.conv2d(x, weight, bias, (sh, sw), (ph, pw))
+    if pt != pb or pl != pr:
+        x = torch.nn.functional.pad(x, (pl, pr, pt, pb))
+        pt = pl = 0
+        y = torch.nn.functional.conv2d(x, weight, bias, (sh, sw), 0)
+    else:
+        y = torch.nn.functional.conv2d(x, weight, bias, (sh, sw), (pt, pl))
     if res is not None:
         y = y + res
     if act == 'relu':
@@ -462,12 +477,27 @@ def conv2d_mod(conv: torch.nn.Module, x: torch.Tensor, act: str = 'none',
     groups, tiny K_out)."""
     w = conv.weight
     kh, kw = w.shape[2], w.shape[3]
+    c = x.shape[1]
     eligible = (x.is_cuda and x.dtype == torch.bfloat16
                 and w.dtype == torch.bfloat16
-                and x.shape[1] % 8 == 0 and w.shape[0] >= 16
+                and w.shape[0] >= 16
                 and conv.dilation == (1, 1) and conv.groups == 1
                 and x.is_contiguous(memory_format=torch.channels_last)
-                and hip_available() and not _env_flag('VFA_NO_CONV'))
+                and hip_available() and not _env_flag('VFA_NO_CONV')
+                and not _env_flag('VFA_FORCE_TORCH_OPS'))
+    if eligible and c % 8 != 0:
+        # stems (C=3/2/1) and the RAFT corr input (C=324): zero-pad the
+        # channel dim — weight padded once and cached on the module, input
+        # padded inside the conv's (fused) pad pass
+        wp = getattr(conv, '_vfa_wpad', None)
+        if wp is None or wp.device != x.device or wp.dtype != x.dtype:
+            c8 = (c + 7) // 8 * 8
+            wp = torch.nn.functional.pad(
+                w, (0, 0, 0, 0, 0, c8 - c)).contiguous(
+                    memory_format=torch.channels_last)
+            conv._vfa_wpad = wp
+        return conv2d_act(x, wp, conv.bias, conv.stride, conv.padding, act,
+                          res)
     if eligible and kh == 1 and kw == 1 and conv.stride == (1, 1) \
             and w.shape[0] % 8 == 0:
         return conv1x1_act(x, w, conv.bias, act, res)

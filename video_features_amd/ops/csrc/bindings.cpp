@@ -52,7 +52,7 @@ void vfa_conv2d_nhwc(const void*, const void*, const void*, const void*,
                      void*, int, int, int, int, int, int, int, int, int,
                      int, int, int, hipStream_t);
 void vfa_pad2d_nhwc(const void*, void*, int, int, int, int, int, int, int,
-                    int, hipStream_t);
+                    int, int, hipStream_t);
 }
 
 namespace {
@@ -421,7 +421,8 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> bias,
                           c10::optional<torch::Tensor> res,
                           int64_t stride_h, int64_t stride_w,
-                          int64_t pad_h, int64_t pad_w, int64_t act) {
+                          int64_t pt, int64_t pb, int64_t pl, int64_t pr,
+                          int64_t act) {
   // x (B, C, H, W) channels_last bf16; w (K, C, KH, KW) channels_last bf16
   // (physical (K, KH, KW, C) = the (N, Kr) B-operand); out (B, K, OH, OW)
   // channels_last.  Implicit-GEMM MFMA kernel; input pre-padded by the
@@ -434,16 +435,20 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
   const int b = (int)x.size(0), c = (int)x.size(1);
   const int h = (int)x.size(2), ww = (int)x.size(3);
   const int kout = (int)w.size(0), kh = (int)w.size(2), kw = (int)w.size(3);
-  TORCH_CHECK(w.size(1) == c && c % 8 == 0, "C % 8 == 0 required");
-  const int hp = h + 2 * (int)pad_h, wp = ww + 2 * (int)pad_w;
+  // C % 8 != 0 (stems, the RAFT corr input): the pad pass widens the
+  // channel dim with zeros; the weight must arrive already zero-padded to
+  // the same c8 (conv2d_mod caches that)
+  const int c8 = (c + 7) / 8 * 8;
+  TORCH_CHECK(w.size(1) == c8, "weight C must be input C padded up to 8");
+  const int hp = h + (int)(pt + pb), wp = ww + (int)(pl + pr);
   const int oh = (hp - kh) / (int)stride_h + 1;
   const int ow = (wp - kw) / (int)stride_w + 1;
   auto stream = current_stream();
   torch::Tensor xp = x;
-  if (pad_h > 0 || pad_w > 0) {
-    xp = torch::empty({b, hp, wp, c}, x.options());
-    vfa_pad2d_nhwc(x.data_ptr(), xp.data_ptr(), b, h, ww, c, (int)pad_h,
-                   (int)pad_h, (int)pad_w, (int)pad_w, stream);
+  if (pt > 0 || pb > 0 || pl > 0 || pr > 0 || c8 != c) {
+    xp = torch::empty({b, hp, wp, c8}, x.options());
+    vfa_pad2d_nhwc(x.data_ptr(), xp.data_ptr(), b, h, ww, c, c8, (int)pt,
+                   (int)pb, (int)pl, (int)pr, stream);
     xp = xp.permute({0, 3, 1, 2});   // logical NCHW view, CL physical
   }
   const void* bptr = nullptr;
@@ -463,20 +468,21 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
   auto out = torch::empty({(long)b, oh, ow, kout}, x.options())
                  .permute({0, 3, 1, 2});
   vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr, out.data_ptr(),
-                  b, hp, wp, c, oh, ow, kout, kh, kw, (int)stride_h,
+                  b, hp, wp, c8, oh, ow, kout, kh, kw, (int)stride_h,
                   (int)stride_w, (int)act, stream);
   return out;
 }
 
 torch::Tensor temporal_merge(torch::Tensor y, int64_t b, int64_t kt,
-                             int64_t st, int64_t p0, bool relu) {
+                             int64_t st, int64_t p0, int64_t p1,
+                             bool relu) {
   // y (B*T, kt*O, H, W) channels_last -> (B*T', O, H, W) channels_last
   TORCH_CHECK(y.is_cuda() && y.dim() == 4);
   TORCH_CHECK(cl_contig(y), "channels_last expected");
   const int bt = (int)y.size(0), cin = (int)y.size(1);
   const int h = (int)y.size(2), w = (int)y.size(3);
   const int o = cin / (int)kt, t = bt / (int)b;
-  const int to = (int)((t + 2 * p0 - kt) / st + 1);
+  const int to = (int)((t + p0 + p1 - kt) / st + 1);
   auto out = torch::empty({(long)b * to, h, w, o}, y.options())
                  .permute({0, 3, 1, 2});
   vfa_temporal_merge(y.data_ptr(), out.data_ptr(), (int)b, t, to, (int)kt,

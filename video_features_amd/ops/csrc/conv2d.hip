@@ -327,18 +327,22 @@ void launch_conv(const void* x, const void* w, const void* bias,
 }
 
 // ------------------------------------------------------------- pad kernel
-// zero-pad H/W of an NHWC tensor: (B, H, W, C) -> (B, H+pt+pb, W+pl+pr, C).
-// Memory-bound; 16-B vectorized over channels (C % 8 == 0).
+// zero-pad H/W and optionally CHANNELS of an NHWC tensor:
+// (B, H, W, Cin) -> (B, H+pt+pb, W+pl+pr, Cout), Cout % 8 == 0, zeros for
+// c >= Cin.  The channel pad is how C=3/C=2 stems (ResNet/RAFT/I3D) and
+// the C=324 RAFT corr input reach the implicit-GEMM kernel (which needs
+// C % 8 == 0).  Memory-bound; 16-B vectorized, elementwise on the one
+// segment straddling Cin.
 __global__ void pad2d_nhwc_kernel(const __bf16* __restrict__ x,
                                   __bf16* __restrict__ out, int b, int h,
-                                  int w, int c, int pt, int pb, int pl,
-                                  int pr) {
+                                  int w, int cin, int cout, int pt, int pb,
+                                  int pl, int pr) {
   const int hp = h + pt + pb, wp = w + pl + pr;
-  const long long total = (long long)b * hp * wp * (c / 8);
+  const long long total = (long long)b * hp * wp * (cout / 8);
   for (long long i = (long long)blockIdx.x * blockDim.x + threadIdx.x;
        i < total; i += (long long)gridDim.x * blockDim.x) {
-    const int cseg = (int)(i % (c / 8));
-    long long t = i / (c / 8);
+    const int cseg = (int)(i % (cout / 8));
+    long long t = i / (cout / 8);
     const int xw = (int)(t % wp);
     t /= wp;
     const int xh = (int)(t % hp);
@@ -346,11 +350,17 @@ __global__ void pad2d_nhwc_kernel(const __bf16* __restrict__ x,
     uint4 v = {0u, 0u, 0u, 0u};
     const int sy = xh - pt, sx = xw - pl;
     if (sy >= 0 && sy < h && sx >= 0 && sx < w) {
-      v = *reinterpret_cast<const uint4*>(
-          x + (((long long)bi * h + sy) * w + sx) * c + cseg * 8);
+      const long long src = (((long long)bi * h + sy) * w + sx) * cin;
+      if (cseg * 8 + 8 <= cin) {
+        v = *reinterpret_cast<const uint4*>(x + src + cseg * 8);
+      } else if (cseg * 8 < cin) {
+        __bf16* e = reinterpret_cast<__bf16*>(&v);
+        for (int j = 0; cseg * 8 + j < cin; ++j)
+          e[j] = x[src + cseg * 8 + j];
+      }
     }
     *reinterpret_cast<uint4*>(
-        out + (((long long)bi * hp + xh) * wp + xw) * c + cseg * 8) = v;
+        out + (((long long)bi * hp + xh) * wp + xw) * cout + cseg * 8) = v;
   }
 }
 
@@ -373,15 +383,16 @@ void vfa_conv2d_nhwc(const void* x, const void* w, const void* bias,
   }
 }
 
-void vfa_pad2d_nhwc(const void* x, void* out, int b, int h, int w, int c,
-                    int pt, int pb, int pl, int pr, hipStream_t stream) {
+void vfa_pad2d_nhwc(const void* x, void* out, int b, int h, int w, int cin,
+                    int cout, int pt, int pb, int pl, int pr,
+                    hipStream_t stream) {
   const long long total = (long long)b * (h + pt + pb) * (w + pl + pr) *
-                          (c / 8);
+                          (cout / 8);
   const int threads = 256;
   const int blocks = (int)min((total + threads - 1) / threads, 2048LL);
   hipLaunchKernelGGL(pad2d_nhwc_kernel, dim3(blocks), dim3(threads), 0,
-                     stream, (const __bf16*)x, (__bf16*)out, b, h, w, c, pt,
-                     pb, pl, pr);
+                     stream, (const __bf16*)x, (__bf16*)out, b, h, w, cin,
+                     cout, pt, pb, pl, pr);
 }
 
 }  // extern "C"
