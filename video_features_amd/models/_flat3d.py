@@ -41,9 +41,13 @@ def cached_cl_weight(mod, name: str, src: torch.Tensor, build):
 
 
 def flatten_time(x: torch.Tensor) -> torch.Tensor:
-    """(B, C, T, H, W) → (B*T, C, H, W) channels_last, one copy."""
+    """(B, C, T, H, W) → (B*T, C, H, W) channels_last, one copy.
+
+    The explicit .contiguous matters: at B=1 the reshape can alias the
+    NCDHW storage and hand back a non-CL VIEW, which silently dropped
+    every downstream op to its strided slow path."""
     b, c, t, h, w = x.shape
-    y = x.permute(0, 2, 3, 4, 1).reshape(b * t, h, w, c)
+    y = x.permute(0, 2, 3, 4, 1).reshape(b * t, h, w, c).contiguous()
     return y.permute(0, 3, 1, 2)
 
 
