@@ -3,6 +3,8 @@
 // torch-header-heavy) TU rarely recompiles.
 #include <torch/extension.h>
 #include <c10/hip/HIPStream.h>
+#include <mutex>
+#include <unordered_map>
 #include <hip/hip_runtime.h>
 
 extern "C" {
@@ -49,8 +51,8 @@ void vfa_linear_act(const void*, const void*, const void*, const void*,
 void vfa_temporal_merge(const void*, void*, int, int, int, int, int, int,
                         int, long long, int, int, hipStream_t);
 void vfa_conv2d_nhwc(const void*, const void*, const void*, const void*,
-                     void*, int, int, int, int, int, int, int, int, int,
-                     int, int, int, hipStream_t);
+                     const void*, void*, int, int, int, int, int, int, int,
+                     int, int, int, int, int, int, int, hipStream_t);
 void vfa_pad2d_nhwc(const void*, void*, int, int, int, int, int, int, int,
                     int, int, hipStream_t);
 }
@@ -444,12 +446,31 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
   const int oh = (hp - kh) / (int)stride_h + 1;
   const int ow = (wp - kw) / (int)stride_w + 1;
   auto stream = current_stream();
+  // C % 8 == 0: the kernel zero-pads inline (validity check + zero-page
+  // redirect in the A staging) — no materialized pad pass.  Channel
+  // padding (stems) still materializes: the straddling 16-B segment
+  // cannot be partially zeroed by a redirected load.
+  const bool inline_pad = (c8 == c);
   torch::Tensor xp = x;
-  if (pt > 0 || pb > 0 || pl > 0 || pr > 0 || c8 != c) {
+  if (!inline_pad && (pt > 0 || pb > 0 || pl > 0 || pr > 0 || c8 != c)) {
     xp = torch::empty({b, hp, wp, c8}, x.options());
     vfa_pad2d_nhwc(x.data_ptr(), xp.data_ptr(), b, h, ww, c, c8, (int)pt,
                    (int)pb, (int)pl, (int)pr, stream);
     xp = xp.permute({0, 3, 1, 2});   // logical NCHW view, CL physical
+  }
+  // 1 KiB zero page for out-of-image glds redirects (per device, cached)
+  static std::unordered_map<int, torch::Tensor> zpages;
+  static std::mutex zp_mu;
+  torch::Tensor zpage;
+  {
+    std::lock_guard<std::mutex> lk(zp_mu);
+    auto it = zpages.find((int)x.get_device());
+    if (it == zpages.end()) {
+      zpage = torch::zeros({512}, x.options());
+      zpages.emplace((int)x.get_device(), zpage);
+    } else {
+      zpage = it->second;
+    }
   }
   const void* bptr = nullptr;
   torch::Tensor bc;
@@ -467,9 +488,16 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
   }
   auto out = torch::empty({(long)b, oh, ow, kout}, x.options())
                  .permute({0, 3, 1, 2});
-  vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr, out.data_ptr(),
-                  b, hp, wp, c8, oh, ow, kout, kh, kw, (int)stride_h,
-                  (int)stride_w, (int)act, stream);
+  if (inline_pad)
+    vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr,
+                    zpage.data_ptr(), out.data_ptr(), b, h, ww, c8, oh, ow,
+                    kout, kh, kw, (int)stride_h, (int)stride_w, (int)pt,
+                    (int)pl, (int)act, stream);
+  else
+    vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr,
+                    zpage.data_ptr(), out.data_ptr(), b, hp, wp, c8, oh,
+                    ow, kout, kh, kw, (int)stride_h, (int)stride_w, 0, 0,
+                    (int)act, stream);
   return out;
 }
 

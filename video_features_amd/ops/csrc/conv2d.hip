@@ -49,48 +49,60 @@ __device__ __forceinline__ float conv_act_f(float x, int kind) {
 }
 
 struct ConvGeom {
-  int b, hp, wp, cin;       // padded input
+  int b, hp, wp, cin;       // input dims as addressed (REAL h/w when the
+                            // kernel pads inline; pre-padded dims when the
+                            // pad was materialized, with pt = pl = 0)
   int oh, ow, kout;         // output
   int kh, kw, sh, sw;       // filter / stride
+  int pt, pl;               // inline zero-pad (top/left); bottom/right are
+                            // implied by oh/ow + the validity check
   int kr;                   // KH*KW*C
 };
 
-// decompose output-pixel index m -> flat element base into x_pad
-__device__ __forceinline__ long long pix_base(const ConvGeom& g, int m) {
+// decompose output-pixel index m -> (image base, receptive-field origin)
+struct PixRef {
+  long long base;           // bi * H * W * C
+  int y0, x0;               // oy*sh - pt, ox*sw - pl (may be negative)
+};
+
+__device__ __forceinline__ PixRef pix_ref(const ConvGeom& g, int m) {
   const int ox = m % g.ow;
   const int t = m / g.ow;
   const int oy = t % g.oh;
   const int bi = t / g.oh;
-  return (((long long)bi * g.hp + (long long)oy * g.sh) * g.wp +
-          (long long)ox * g.sw) * g.cin;
+  return PixRef{(long long)bi * g.hp * g.wp * g.cin,
+                oy * g.sh - g.pt, ox * g.sw - g.pl};
 }
 
-// k (reduction index) -> element offset into x_pad relative to pix_base
-__device__ __forceinline__ long long tap_off(const ConvGeom& g, int k) {
+// Stage the (ROWS x 64) A-tile via glds: per-thread row refs precomputed;
+// the per-lane k position picks tap (r, s) and channel; out-of-image taps
+// redirect the load to a zero page (the inline zero-padding — no
+// materialized pad pass).  piece splits the wave's glds issues across MFMA
+// half-steps (-1 = all).
+template <int ROWS, int WAVES>
+__device__ __forceinline__ void conv_stage_a(
+    const __bf16* __restrict__ x, const __bf16* __restrict__ zp,
+    const ConvGeom& g, const PixRef* __restrict__ aref, int k0,
+    char* lds_base, int wave, int lane, int kfrac, int piece, int npieces) {
+  constexpr int NSUB = ROWS * 128 / 1024;
+  constexpr int PER_WAVE = NSUB / WAVES;
+  const int k = k0 + kfrac;
   const int tap = k / g.cin;            // uniform-cost u32 div
   const int c = k - tap * g.cin;
   const int r = tap / g.kw;
   const int s = tap - r * g.kw;
-  return (long long)(r * g.wp + s) * g.cin + c;
-}
-
-// Stage the (ROWS x 64) A-tile via glds: per-thread row bases precomputed
-// in abase[]; the per-lane k position adds the tap offset for this K-tile.
-// piece splits the wave's glds issues across MFMA half-steps (-1 = all).
-template <int ROWS, int WAVES>
-__device__ __forceinline__ void conv_stage_a(
-    const __bf16* __restrict__ x, const ConvGeom& g,
-    const long long* __restrict__ abase, int k0, char* lds_base, int wave,
-    int lane, int kfrac, int piece, int npieces) {
-  constexpr int NSUB = ROWS * 128 / 1024;
-  constexpr int PER_WAVE = NSUB / WAVES;
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
     if (piece >= 0 && (i * npieces) / PER_WAVE != piece) continue;
     const int sub = wave * PER_WAVE + i;
-    const long long src_elem = abase[i] + tap_off(g, k0 + kfrac);
+    const int iy = aref[i].y0 + r, ix = aref[i].x0 + s;
+    const bool ok = (unsigned)iy < (unsigned)g.hp &&
+                    (unsigned)ix < (unsigned)g.wp;
+    const __bf16* src =
+        ok ? x + aref[i].base + ((long long)iy * g.wp + ix) * g.cin + c
+           : zp + (lane & 63) * 8;
     __builtin_amdgcn_global_load_lds(
-        reinterpret_cast<const unsigned int*>(x + src_elem),
+        reinterpret_cast<const unsigned int*>(src),
         reinterpret_cast<unsigned int*>(lds_base + sub * 1024), 16, 0, 0);
   }
 }
@@ -119,7 +131,7 @@ __device__ __forceinline__ void conv_stage_w(
 }
 
 // Bounds-checked A staging (M-edge tiles and the ragged last K-tile):
-// same LDS image, zero fill.
+// same LDS image, zero fill (covers out-of-image taps too).
 template <int ROWS, int THREADS>
 __device__ __forceinline__ void conv_guard_a(
     const __bf16* __restrict__ x, const ConvGeom& g, int m0, int mtotal,
@@ -130,8 +142,15 @@ __device__ __forceinline__ void conv_guard_a(
     const int m = m0 + row;
     const int k = k0 + seg * 8;
     if (m < mtotal && k < g.kr) {
-      const long long src = pix_base(g, m) + tap_off(g, k);
-      v = *reinterpret_cast<const uint4*>(x + src);
+      const PixRef pr = pix_ref(g, m);
+      const int tap = k / g.cin;
+      const int c = k - tap * g.cin;
+      const int r = tap / g.kw;
+      const int s = tap - r * g.kw;
+      const int iy = pr.y0 + r, ix = pr.x0 + s;
+      if ((unsigned)iy < (unsigned)g.hp && (unsigned)ix < (unsigned)g.wp)
+        v = *reinterpret_cast<const uint4*>(
+            x + pr.base + ((long long)iy * g.wp + ix) * g.cin + c);
     }
     *reinterpret_cast<uint4*>(lds_base + cswz(row * 128 + seg * 16)) = v;
   }
@@ -161,6 +180,7 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
                         const __bf16* __restrict__ wgt,
                         const __bf16* __restrict__ bias,
                         const __bf16* __restrict__ res,
+                        const __bf16* __restrict__ zp,
                         __bf16* __restrict__ out, ConvGeom g, int mtotal,
                         int tiles_m, int tiles_n) {
   constexpr int MI = BM / (WAVES / WN) / 16;   // 16-row MFMA tiles / wave
@@ -194,24 +214,24 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
   const bool tile_full = valid_m >= BM && valid_n >= BN;
   const bool kfull = (kr % BK) == 0;
 
-  // per-thread A row bases for the glds path (same rows every K-tile);
+  // per-thread A row refs for the glds path (same rows every K-tile);
   // the per-lane swizzled byte offset contributes kfrac elements
   const int off_log = (lane * 16) ^ ((((lane * 16) >> 9) & 1) << 5);
   const int a_rin = off_log >> 7;
   const int kfrac = (off_log & 127) >> 1;
-  long long abase[PER_WAVE];
+  PixRef aref[PER_WAVE];
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
     const int sub = wave * PER_WAVE + i;
     const int m = m0 + sub * 8 + a_rin;
-    abase[i] = pix_base(g, m < mtotal ? m : mtotal - 1);
+    aref[i] = pix_ref(g, m < mtotal ? m : mtotal - 1);
   }
 
   auto stage = [&](int buf, int kt) {
     const int k0 = kt * BK;
     if (tile_full && (kfull || kt + 1 < nk)) {
-      conv_stage_a<BM, WAVES>(x, g, abase, k0, sA(buf), wave, lane, kfrac,
-                              -1, 1);
+      conv_stage_a<BM, WAVES>(x, zp, g, aref, k0, sA(buf), wave, lane,
+                              kfrac, -1, 1);
       conv_stage_w<BN, WAVES>(wgt + (long long)n0 * kr + k0, kr, sB(buf),
                               wave, lane, -1, 1);
     } else {
@@ -252,7 +272,7 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
             sA(cur) + cswz(arow * 128 + kk * 64 + hi4 * 16));
       }
       if (glds_nxt) {
-        conv_stage_a<BM, WAVES>(x, g, abase, k0_nxt, sA(1 - cur), wave,
+        conv_stage_a<BM, WAVES>(x, zp, g, aref, k0_nxt, sA(1 - cur), wave,
                                 lane, kfrac, kk, 2);
         conv_stage_w<BN, WAVES>(wgt + (long long)n0 * kr + k0_nxt, kr,
                                 sB(1 - cur), wave, lane, kk, 2);
@@ -289,22 +309,22 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
 
 template <int ACT, int BM, int BN, int WAVES, int WN>
 void launch_cfg(const void* x, const void* w, const void* bias,
-                const void* res, void* out, const ConvGeom& g, int m,
-                hipStream_t stream) {
+                const void* res, const void* zp, void* out,
+                const ConvGeom& g, int m, hipStream_t stream) {
   const int tiles_m = (m + BM - 1) / BM, tiles_n = (g.kout + BN - 1) / BN;
   const dim3 grid(tiles_m * tiles_n);
   const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
   hipLaunchKernelGGL((conv2d_nhwc_kernel<ACT, BM, BN, WAVES, WN>), grid,
                      dim3(WAVES * 64), lds, stream, (const __bf16*)x,
                      (const __bf16*)w, (const __bf16*)bias,
-                     (const __bf16*)res, (__bf16*)out, g, m, tiles_m,
-                     tiles_n);
+                     (const __bf16*)res, (const __bf16*)zp, (__bf16*)out,
+                     g, m, tiles_m, tiles_n);
 }
 
 template <int ACT>
 void launch_conv(const void* x, const void* w, const void* bias,
-                 const void* res, void* out, const ConvGeom& g,
-                 hipStream_t stream) {
+                 const void* res, const void* zp, void* out,
+                 const ConvGeom& g, hipStream_t stream) {
   const int m = g.b * g.oh * g.ow;
   // pick the tile minimizing padded work, preferring the bigger tile when
   // waste ties and the grid still fills the chip (256 CUs)
@@ -319,11 +339,11 @@ void launch_conv(const void* x, const void* w, const void* bias,
   const long long c128 = cost(128, 128, 120);
   const long long c64 = cost(256, 64, 120);
   if (c256 <= c128 && c256 <= c64)
-    launch_cfg<ACT, 256, 256, 8, 4>(x, w, bias, res, out, g, m, stream);
+    launch_cfg<ACT, 256, 256, 8, 4>(x, w, bias, res, zp, out, g, m, stream);
   else if (c64 < c128)
-    launch_cfg<ACT, 256, 64, 4, 1>(x, w, bias, res, out, g, m, stream);
+    launch_cfg<ACT, 256, 64, 4, 1>(x, w, bias, res, zp, out, g, m, stream);
   else
-    launch_cfg<ACT, 128, 128, 4, 2>(x, w, bias, res, out, g, m, stream);
+    launch_cfg<ACT, 128, 128, 4, 2>(x, w, bias, res, zp, out, g, m, stream);
 }
 
 // ------------------------------------------------------------- pad kernel
@@ -369,17 +389,21 @@ __global__ void pad2d_nhwc_kernel(const __bf16* __restrict__ x,
 extern "C" {
 
 // act: 0 none, 1 relu, 2 quick_gelu, 3 gelu_tanh, 4 leaky_relu(0.1)
+// pt/pl: inline zero-pad applied by the kernel (zp = 1 KiB zero page);
+// pass 0 with pre-padded h/w when the pad was materialized
 void vfa_conv2d_nhwc(const void* x, const void* w, const void* bias,
-                     const void* res, void* out, int b, int hp, int wp,
-                     int cin, int oh, int ow, int kout, int kh, int kw,
-                     int sh, int sw, int act, hipStream_t stream) {
-  ConvGeom g{b, hp, wp, cin, oh, ow, kout, kh, kw, sh, sw, kh * kw * cin};
+                     const void* res, const void* zp, void* out, int b,
+                     int hp, int wp, int cin, int oh, int ow, int kout,
+                     int kh, int kw, int sh, int sw, int pt, int pl,
+                     int act, hipStream_t stream) {
+  ConvGeom g{b, hp, wp, cin, oh, ow, kout, kh, kw, sh, sw, pt, pl,
+             kh * kw * cin};
   switch (act) {
-    case 0: launch_conv<0>(x, w, bias, res, out, g, stream); break;
-    case 1: launch_conv<1>(x, w, bias, res, out, g, stream); break;
-    case 2: launch_conv<2>(x, w, bias, res, out, g, stream); break;
-    case 3: launch_conv<3>(x, w, bias, res, out, g, stream); break;
-    case 4: launch_conv<4>(x, w, bias, res, out, g, stream); break;
+    case 0: launch_conv<0>(x, w, bias, res, zp, out, g, stream); break;
+    case 1: launch_conv<1>(x, w, bias, res, zp, out, g, stream); break;
+    case 2: launch_conv<2>(x, w, bias, res, zp, out, g, stream); break;
+    case 3: launch_conv<3>(x, w, bias, res, zp, out, g, stream); break;
+    case 4: launch_conv<4>(x, w, bias, res, zp, out, g, stream); break;
   }
 }
 
