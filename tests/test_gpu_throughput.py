@@ -45,14 +45,12 @@ def test_clip_throughput_floor():
 
 
 def test_i3d_raft_throughput_floor():
-    run_bench('i3d_raft', 12)
+    run_bench('i3d_raft', 18)
 
 
 def test_resnet50_throughput_floor():
-    run_bench('resnet50', 9_000)
+    run_bench('resnet50', 15_000)
 
 
 def test_vggish_r21d_throughput_floor():
-    # R(2+1)D-34 (BASELINE's depth) is ~1.7x the -18 FLOPs; floor set from
-    # the -18 measurement scaled down accordingly
-    run_bench('vggish_r21d', 250, extra=('--clips-per-step', '8'))
+    run_bench('vggish_r21d', 400)
