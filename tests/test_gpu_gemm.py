@@ -73,6 +73,44 @@ def test_linear_thin_streaming(dev, m, n, k, act):
     assert rel < 3e-2, rel
 
 
+@pytest.mark.parametrize('m,n,k', [
+    (19200, 2304, 768),   # CLIP qkv (fb384) — the 8-phase pipelined path
+    (19200, 768, 3072),   # CLIP fc2
+    (4096, 2048, 768),
+    (4096, 256, 192),     # minimum K depth (3 tiles)
+])
+@pytest.mark.parametrize('act', ['none', 'quick_gelu'])
+def test_linear_8phase(dev, m, n, k, act):
+    """The deep-pipelined 256^2 kernel (full tiles, counted-vmcnt raw
+    barriers); numerics vs fp32 torch."""
+    ops = _hip_loaded()
+    torch.manual_seed(0)
+    x = (torch.randn(m, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) / (k ** 0.25)).to(torch.bfloat16)
+    b = torch.randn(n, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, b, act)
+    ref = _ref(x, w, b, act)
+    err = (out.float() - ref).abs()
+    rel = err.max().item() / max(ref.abs().max().item(), 1e-6)
+    assert rel < 3e-2, rel
+    assert (err.mean() / ref.abs().mean().clamp_min(1e-6)).item() < 5e-3
+
+
+def test_linear_8phase_residual(dev):
+    """fc2 + residual add fused in the 8-phase epilogue."""
+    ops = _hip_loaded()
+    torch.manual_seed(4)
+    m, n, k = 19200, 768, 3072
+    x = (torch.randn(m, k, device=dev) / 8).to(torch.bfloat16)
+    w = (torch.randn(n, k, device=dev) / 8).to(torch.bfloat16)
+    b = torch.randn(n, device=dev).to(torch.bfloat16)
+    r = torch.randn(m, n, device=dev).to(torch.bfloat16)
+    out = ops.linear_act(x, w, b, 'none', r)
+    ref = _ref(x, w, b, 'none') + r.float()
+    rel = (out.float() - ref).abs().max().item() / ref.abs().max().item()
+    assert rel < 3e-2, rel
+
+
 def test_linear_thin_residual(dev):
     """bottleneck conv3 epilogue: bias + residual + relu on the thin path."""
     ops = _hip_loaded()

@@ -18,6 +18,7 @@
 // Partial tiles (M/N/K tails) take a bounds-checked vector-staging path
 // with an identical LDS image.
 #include "vfa_common.h"
+#include <cstdlib>
 
 typedef __bf16 bf16x8 __attribute__((ext_vector_type(8)));
 typedef float f32x4 __attribute__((ext_vector_type(4)));
@@ -234,6 +235,210 @@ void linear_act_kernel(const __bf16* __restrict__ a,
   }
 }
 
+// --------------------------------------------------- deep-pipelined 256^2
+// The 8-phase 256x256 structure (cdna_hip_programming.md §5 template): one
+// phase per C-quadrant, 16 MFMA per phase, TWO glds per phase, raw
+// s_barrier with a counted vmcnt(6) — staged loads stay in flight ACROSS
+// barriers (the plain-__syncthreads 2-buffer loop drains vmcnt(0) at every
+// barrier; that stall was ~20% per the guide).
+//
+// Provably-correct schedule with uniform 2 glds/phase:
+//  - A ring: 2 tiles x 4 quadrant blocks of 8 KiB.  Block q holds the 64
+//    rows quadrant q reads (rows q*32..+32 and 128+q*32..+32).  A(T+1, q)
+//    is staged at phase (T, q) -> stage-to-use distance EXACTLY 4 phases.
+//  - B ring: 3 tiles x 4 quarter blocks of 8 KiB (quarter h = tile cols
+//    h*64..+64 = wave wn=h's fragment rows).  B(T+2, q) staged at phase
+//    (T, q) -> distance 5..8 phases.
+//  - vmcnt(6) leaves at most the newest 3 phases' stages (2 glds each) in
+//    flight, so anything >= 4 phases old has landed.  Past the end of K
+//    the stages keep issuing into dead blocks (source redirected to tile
+//    0) so the rate stays uniform and the invariant holds at ramp-down.
+// FULL tiles only (m%256==0, n%256==0, k%64==0) — no edge guards anywhere.
+template <int ACT>
+__global__ __launch_bounds__(512, 1)
+void linear8p_kernel(const __bf16* __restrict__ a,
+                     const __bf16* __restrict__ w,
+                     const __bf16* __restrict__ bias,
+                     const __bf16* __restrict__ res,
+                     __bf16* __restrict__ c, int m, int n, int k,
+                     int tiles_m, int tiles_n) {
+  extern __shared__ __attribute__((aligned(1024))) char smem8[];
+  const int nk = k >> 6;
+
+  const int nwg = tiles_m * tiles_n;
+  int wg = blockIdx.x;
+  {
+    const int xcd = wg % 8, orig = wg / 8;
+    const int q = nwg / 8, r = nwg % 8;
+    wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q) + orig;
+  }
+  const int tile_n = wg / tiles_m, tile_m = wg % tiles_m;
+  const int m0 = tile_m * 256, n0 = tile_n * 256;
+
+  const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
+  const int lo = lane & 15, hi4 = lane >> 4;
+  const int wm = wave >> 2, wn = wave & 3;
+
+  auto LA = [&](int t, int q) { return smem8 + (t * 4 + q) * 8192; };
+  auto LB = [&](int t, int h) {
+    return smem8 + 65536 + (t * 4 + h) * 8192;
+  };
+
+  // per-wave glds geometry: each wave writes 1 KiB of an 8 KiB block
+  // (lane-linear dest); the st_16x32 swizzle rides the SOURCE address
+  const int off = (wave * 1024 + lane * 16);
+  const int off_log = off ^ (((off >> 9) & 1) << 5);
+  const int rib = off_log >> 7;            // row in block (0..63)
+  const int kfrac = (off_log & 127) >> 1;  // k elems within the K-tile
+
+  // A: block q, row-in-block rib -> global row
+  const int a_row_off = rib < 32 ? rib : 96 + rib;  // +128-32 for wm=1 half
+  auto stage_a = [&](int T, int q) {
+    // beyond-K stages read tile 0 into the dead block (rate uniformity)
+    const int kt = T < nk ? T : 0;
+    const __bf16* src =
+        a + (long long)(m0 + q * 32 + a_row_off) * k + kt * 64 + kfrac;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int*>(src),
+        reinterpret_cast<unsigned int*>(LA(T & 1, q) + wave * 1024), 16, 0,
+        0);
+  };
+  auto stage_b = [&](int T, int h) {
+    const int kt = T < nk ? T : 0;
+    const __bf16* src =
+        w + (long long)(n0 + h * 64 + rib) * k + kt * 64 + kfrac;
+    __builtin_amdgcn_global_load_lds(
+        reinterpret_cast<const unsigned int*>(src),
+        reinterpret_cast<unsigned int*>(LB(T % 3, h) + wave * 1024), 16, 0,
+        0);
+  };
+
+  // prologue: A(0).q0-3, B(0).h0-3, B(1).h0-3 — 12 glds per wave; wait
+  // the first 8 (A0 + B0), leave B1 in flight
+  for (int q = 0; q < 4; ++q) stage_a(0, q);
+  for (int h = 0; h < 4; ++h) stage_b(0, h);
+  for (int h = 0; h < 4; ++h) stage_b(1, h);
+  // s_waitcnt imm: vmcnt [3:0]+[15:14], expcnt [6:4], lgkmcnt [13:8]
+  __builtin_amdgcn_s_waitcnt(0x3f74);      // vmcnt(4) only
+  __builtin_amdgcn_s_barrier();
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = f32x4{0.f, 0.f, 0.f, 0.f};
+
+  for (int T = 0; T < nk; ++T) {
+    char* la_t = LA(T & 1, 0);
+    char* lb_t = LB(T % 3, wn);
+#pragma unroll
+    for (int q = 0; q < 4; ++q) {            // one C-quadrant per phase
+      // fragments for this phase (hipcc inserts the lgkm ladders)
+      bf16x8 afr[2], bfr[4][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk) {
+#pragma unroll
+        for (int j = 0; j < 4; ++j) {
+          const int brow = j * 16 + lo;
+          bfr[j][kk] = *reinterpret_cast<const bf16x8*>(
+              lb_t + swz(brow * 128 + kk * 64 + hi4 * 16));
+        }
+      }
+#pragma unroll
+      for (int mi2 = 0; mi2 < 2; ++mi2) {
+        const int arow = wm * 32 + mi2 * 16 + lo;
+        afr[mi2] = *reinterpret_cast<const bf16x8*>(
+            la_t + q * 8192 + swz(arow * 128 + 0 * 64 + hi4 * 16));
+      }
+      // the second kk A-fragments read after the first MFMA batch would
+      // serialize; read both up front (4 reads total)
+      bf16x8 afr1[2];
+#pragma unroll
+      for (int mi2 = 0; mi2 < 2; ++mi2) {
+        const int arow = wm * 32 + mi2 * 16 + lo;
+        afr1[mi2] = *reinterpret_cast<const bf16x8*>(
+            la_t + q * 8192 + swz(arow * 128 + 64 + hi4 * 16));
+      }
+      // stage: A(T+1, q) and B(T+2, q) — 2 glds, uniform every phase
+      stage_a(T + 1, q);
+      stage_b(T + 2, q);
+      __builtin_amdgcn_s_waitcnt(0x3f76);    // vmcnt(6): 3 phases in flight
+      __builtin_amdgcn_s_barrier();
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mi2 = 0; mi2 < 2; ++mi2)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mi2], bfr[j][0], acc[q * 2 + mi2][j], 0, 0, 0);
+#pragma unroll
+      for (int mi2 = 0; mi2 < 2; ++mi2)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr1[mi2], bfr[j][1], acc[q * 2 + mi2][j], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+  }
+
+  // epilogue: LDS-bounce to 16-B vectorized bias/res/store (full tiles,
+  // no guards).  __syncthreads drains the dummy glds still in flight.
+  __syncthreads();
+  __bf16* ep = reinterpret_cast<__bf16*>(smem8) + wave * (16 * 72);
+#pragma unroll
+  for (int mi = 0; mi < 8; ++mi) {
+#pragma unroll
+    for (int jj = 0; jj < 4; ++jj)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        ep[(hi4 * 4 + r) * 72 + jj * 16 + lo] = (__bf16)acc[mi][jj][r];
+    __builtin_amdgcn_s_waitcnt(0xc07f);      // lgkmcnt(0), wave-local tile
+#pragma unroll
+    for (int p = 0; p < 2; ++p) {
+      const int er = p * 8 + (lane >> 3);
+      const int ec = (lane & 7) * 8;
+      const long long row = m0 + wm * 128 + mi * 16 + er;
+      const int col = n0 + wn * 64 + ec;
+      bf16x8 v8 = *reinterpret_cast<const bf16x8*>(ep + er * 72 + ec);
+      bf16x8 o8;
+      bf16x8 b8{}, rr8{};
+      if (bias) b8 = *reinterpret_cast<const bf16x8*>(bias + col);
+      if (res) rr8 = *reinterpret_cast<const bf16x8*>(res + row * n + col);
+#pragma unroll
+      for (int e = 0; e < 8; ++e) {
+        float v = (float)v8[e];
+        if (bias) v += (float)b8[e];
+        if (res) v += (float)rr8[e];
+        o8[e] = (__bf16)act_f(v, ACT);
+      }
+      *reinterpret_cast<bf16x8*>(c + row * n + col) = o8;
+    }
+    __builtin_amdgcn_s_waitcnt(0xc07f);      // reads done before reuse
+  }
+}
+
+template <int ACT>
+bool launch_8p(const void* a, const void* w, const void* bias,
+               const void* res, void* c, int m, int n, int k,
+               hipStream_t stream) {
+  if (m % 256 || n % 256 || k % 64 || k / 64 < 3) return false;
+  const long long tiles = (long long)(m / 256) * (n / 256);
+  if (tiles < 64) return false;              // chip fill
+  static bool attr_set[4] = {};
+  if (!attr_set[ACT]) {
+    hipFuncSetAttribute(
+        reinterpret_cast<const void*>(&linear8p_kernel<ACT>),
+        hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
+    attr_set[ACT] = true;
+  }
+  hipLaunchKernelGGL((linear8p_kernel<ACT>), dim3((unsigned)tiles),
+                     dim3(512), 160 * 1024, stream, (const __bf16*)a,
+                     (const __bf16*)w, (const __bf16*)bias,
+                     (const __bf16*)res, (__bf16*)c, m, n, k, m / 256,
+                     n / 256);
+  return true;
+}
+
 // ------------------------------------------------------- thin-K streaming
 // M-huge / K-shallow GEMMs (the ResNet/I3D 1x1 convs as GEMM: M = B*H*W up
 // to ~1.2M rows, K = 64..256) are HBM-streaming problems: the tiled
@@ -437,7 +642,12 @@ void launch_linear(const void* a, const void* w, const void* bias,
   // tail tile and the block-round quantization cost more than the smaller
   // tile's overhead — measured 339 vs 528 TF at M=9600) and the grid
   // still fills the chip
+  // 8-phase pipelined kernel: correct but currently BELOW the 2-buffer
+  // structure (1018 vs 1149 TF @8k — the fine per-phase interleave is
+  // still missing, cdna guide §5.5); opt-in via VFA_8P for experiments
+  static const bool use8p = getenv("VFA_8P") != nullptr;
   if (launch_thin<ACT>(a, w, bias, res, c, m, n, k, stream)) return;
+  if (use8p && launch_8p<ACT>(a, w, bias, res, c, m, n, k, stream)) return;
   const long long tiles = (long long)((m + 255) / 256) * ((n + 255) / 256);
   if (n >= 256 && m % 256 == 0 && tiles >= 150)
     launch_tile<ACT, true>(a, w, bias, res, c, m, n, k, stream);
