@@ -331,34 +331,30 @@ void linear8p_kernel(const __bf16* __restrict__ a,
   for (int T = 0; T < nk; ++T) {
     char* la_t = LA(T & 1, 0);
     char* lb_t = LB(T % 3, wn);
+    // B fragments are quadrant-invariant: read ONCE per K-tile (phase 0)
+    // and hold in registers — halves the LDS read traffic per tile
+    bf16x8 bfr[4][2];
 #pragma unroll
     for (int q = 0; q < 4; ++q) {            // one C-quadrant per phase
-      // fragments for this phase (hipcc inserts the lgkm ladders)
-      bf16x8 afr[2], bfr[4][2];
+      if (q == 0) {
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk) {
+        for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-        for (int j = 0; j < 4; ++j) {
-          const int brow = j * 16 + lo;
-          bfr[j][kk] = *reinterpret_cast<const bf16x8*>(
-              lb_t + swz(brow * 128 + kk * 64 + hi4 * 16));
+          for (int j = 0; j < 4; ++j) {
+            const int brow = j * 16 + lo;
+            bfr[j][kk] = *reinterpret_cast<const bf16x8*>(
+                lb_t + swz(brow * 128 + kk * 64 + hi4 * 16));
+          }
+      }
+      bf16x8 afr[2][2];
+#pragma unroll
+      for (int kk = 0; kk < 2; ++kk)
+#pragma unroll
+        for (int mi2 = 0; mi2 < 2; ++mi2) {
+          const int arow = wm * 32 + mi2 * 16 + lo;
+          afr[mi2][kk] = *reinterpret_cast<const bf16x8*>(
+              la_t + q * 8192 + swz(arow * 128 + kk * 64 + hi4 * 16));
         }
-      }
-#pragma unroll
-      for (int mi2 = 0; mi2 < 2; ++mi2) {
-        const int arow = wm * 32 + mi2 * 16 + lo;
-        afr[mi2] = *reinterpret_cast<const bf16x8*>(
-            la_t + q * 8192 + swz(arow * 128 + 0 * 64 + hi4 * 16));
-      }
-      // the second kk A-fragments read after the first MFMA batch would
-      // serialize; read both up front (4 reads total)
-      bf16x8 afr1[2];
-#pragma unroll
-      for (int mi2 = 0; mi2 < 2; ++mi2) {
-        const int arow = wm * 32 + mi2 * 16 + lo;
-        afr1[mi2] = *reinterpret_cast<const bf16x8*>(
-            la_t + q * 8192 + swz(arow * 128 + 64 + hi4 * 16));
-      }
       // stage: A(T+1, q) and B(T+2, q) — 2 glds, uniform every phase
       stage_a(T + 1, q);
       stage_b(T + 2, q);
@@ -366,18 +362,15 @@ void linear8p_kernel(const __bf16* __restrict__ a,
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int mi2 = 0; mi2 < 2; ++mi2)
+      for (int kk = 0; kk < 2; ++kk)
 #pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr[mi2], bfr[j][0], acc[q * 2 + mi2][j], 0, 0, 0);
+        for (int mi2 = 0; mi2 < 2; ++mi2)
 #pragma unroll
-      for (int mi2 = 0; mi2 < 2; ++mi2)
-#pragma unroll
-        for (int j = 0; j < 4; ++j)
-          acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-              afr1[mi2], bfr[j][1], acc[q * 2 + mi2][j], 0, 0, 0);
+          for (int j = 0; j < 4; ++j)
+            acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                afr[mi2][kk], bfr[j][kk], acc[q * 2 + mi2][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();          // phase lockstep
     }
   }
 
@@ -426,7 +419,7 @@ bool launch_8p(const void* a, const void* w, const void* bias,
   if (tiles < 64) return false;              // chip fill
   static bool attr_set[4] = {};
   if (!attr_set[ACT]) {
-    hipFuncSetAttribute(
+    (void)hipFuncSetAttribute(
         reinterpret_cast<const void*>(&linear8p_kernel<ACT>),
         hipFuncAttributeMaxDynamicSharedMemorySize, 160 * 1024);
     attr_set[ACT] = true;
