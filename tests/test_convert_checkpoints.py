@@ -139,3 +139,26 @@ def test_vggish_scheme_roundtrip():
     m2.load_state_dict(back, strict=False)
     for k, v in m2.state_dict().items():
         assert torch.equal(v, sd[k]), k
+
+
+def test_weights_url_fetch_and_load(tmp_path, monkeypatch):
+    """--weights_path accepts a URL: fetched once into the cache, then
+    loaded through the usual converter path (file:// here; http in prod)."""
+    from video_features_amd.models.clip_vit import VisionTransformer
+    from video_features_amd.utils import weights as W
+    from video_features_amd.extractors.base import BaseExtractor
+    torch.manual_seed(0)
+    m = VisionTransformer()
+    ck = tmp_path / 'visual.pt'
+    torch.save(m.state_dict(), str(ck))
+    monkeypatch.setenv('VFA_WEIGHTS_CACHE', str(tmp_path / 'cache'))
+    url = ck.as_uri()
+    assert W.is_url(url) and not W.is_url(str(ck))
+    m2 = VisionTransformer()
+    BaseExtractor.load_weights(m2, url)
+    for k, v in m2.state_dict().items():
+        assert torch.equal(v, m.state_dict()[k]), k
+    # second resolve hits the cache (delete the source to prove it)
+    cached = W.resolve_weights_path(url)
+    ck.unlink()
+    assert W.resolve_weights_path(url) == cached

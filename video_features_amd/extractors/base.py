@@ -94,12 +94,15 @@ class BaseExtractor(torch.nn.Module):
 
     @staticmethod
     def load_weights(model: torch.nn.Module, path: str) -> None:
-        """Load a state-dict file, accepting the published reference
-        checkpoint schemes (OpenAI CLIP / reference i3d / torchvision
-        r2plus1d / torchvggish / DataParallel 'module.' prefixes) via
-        utils.convert_checkpoints.convert_auto."""
+        """Load a state-dict file or URL (http/https/file — downloaded
+        once into the weights cache, utils/weights.py), accepting the
+        published reference checkpoint schemes (OpenAI CLIP / reference
+        i3d / torchvision r2plus1d / torchvggish / sniklaus PWC /
+        DataParallel 'module.' prefixes) via convert_auto."""
         from ..utils.convert_checkpoints import convert_auto
-        sd = torch.load(path, map_location='cpu', weights_only=True)
+        from ..utils.weights import resolve_weights_path
+        sd = torch.load(resolve_weights_path(path), map_location='cpu',
+                        weights_only=True)
         if isinstance(sd, dict) and 'state_dict' in sd \
                 and isinstance(sd['state_dict'], dict):
             sd = sd['state_dict']
