@@ -24,8 +24,17 @@ from torch import nn
 from .. import ops
 
 
+class _ConvLeaky(nn.Sequential):
+    """conv + LeakyReLU(0.1) with the Sequential key scheme (``0.weight``)
+    preserved; on GPU the pair runs as ONE fused implicit-GEMM kernel
+    (act=leaky_relu) via conv2d_mod — dilated refiner convs fall back."""
+
+    def forward(self, x):
+        return ops.conv2d_mod(self[0], x, 'leaky_relu')
+
+
 def _conv(in_ch: int, out_ch: int, stride: int = 1, dilation: int = 1):
-    return nn.Sequential(
+    return _ConvLeaky(
         nn.Conv2d(in_ch, out_ch, 3, stride, padding=dilation, dilation=dilation),
         nn.LeakyReLU(0.1, inplace=True))
 
@@ -48,6 +57,11 @@ class PyramidExtractor(nn.Module):
         self.levels = nn.ModuleList(levels)
 
     def forward(self, x: torch.Tensor) -> List[torch.Tensor]:
+        if x.is_cuda and x.dtype == torch.bfloat16:
+            # channels_last through the pyramid so the fused conv kernel
+            # runs; the decoder's correlation kernel wants NCHW, its
+            # .contiguous() converts per level
+            x = x.contiguous(memory_format=torch.channels_last)
         feats = []
         for level in self.levels:
             x = level(x)
