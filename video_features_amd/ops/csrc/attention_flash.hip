@@ -63,11 +63,17 @@ __device__ __forceinline__ void stage_rows(const __bf16* g, int rows,
 }
 
 // stage V transposed: global rows are keys (64 x 64), LDS is [d][key].
+// Lane mapping is KEY-major: a wave's 64 lanes cover 64 distinct keys at
+// one d-segment, so the transposing u16 scatter touches all 32 LDS banks
+// (~2-way) instead of 4 (16-way with the seg-major mapping — PMC showed
+// 3.5 conflict cycles per LDS instruction on this store).  The global
+// reads lose lane-contiguity but V is L2-hot (the qkv projection just
+// wrote it).
 __device__ __forceinline__ void stage_vt(const __bf16* g, long long row_stride,
                                          __bf16* lds, int nvalid) {
   const int tid = threadIdx.x;
   for (int t = tid; t < KVBLK * 8; t += 256) {
-    const int key = t >> 3, seg = t & 7;
+    const int key = t & 63, seg = t >> 6;
     uint4 v = {0u, 0u, 0u, 0u};
     if (key < nvalid)
       v = *reinterpret_cast<const uint4*>(g + key * row_stride + seg * 8);
