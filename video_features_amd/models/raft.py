@@ -162,7 +162,9 @@ class FlowHead(nn.Module):
         self.conv2 = nn.Conv2d(hidden, 2, 3, 1, 1)
 
     def forward(self, x):
-        return self.conv2(ops.conv2d_mod(self.conv1, x, 'relu'))
+        # conv2 has N=2 outputs — routed via the padded-N path
+        return ops.conv2d_mod(self.conv2,
+                              ops.conv2d_mod(self.conv1, x, 'relu'))
 
 
 class SepConvGRU(nn.Module):

@@ -480,11 +480,26 @@ def conv2d_mod(conv: torch.nn.Module, x: torch.Tensor, act: str = 'none',
     c = x.shape[1]
     eligible = (x.is_cuda and x.dtype == torch.bfloat16
                 and w.dtype == torch.bfloat16
-                and w.shape[0] >= 16
                 and conv.dilation == (1, 1) and conv.groups == 1
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and hip_available() and not _env_flag('VFA_NO_CONV')
                 and not _env_flag('VFA_FORCE_TORCH_OPS'))
+    if eligible and w.shape[0] < 16 and res is None:
+        # tiny-N heads (RAFT flow head N=2, per GRU iteration): pad the
+        # OUTPUT channels to 16 (cached), run in-tree, slice back
+        kout = w.shape[0]
+        wp = getattr(conv, '_vfa_wnpad', None)
+        if wp is None or wp.device != x.device or wp.dtype != x.dtype:
+            wp = torch.nn.functional.pad(
+                w, (0, 0, 0, 0, 0, 0, 0, 16 - kout)).contiguous(
+                    memory_format=torch.channels_last)
+            conv._vfa_wnpad = wp
+            conv._vfa_bnpad = (torch.nn.functional.pad(
+                conv.bias, (0, 16 - kout)) if conv.bias is not None
+                else None)
+        y = conv2d_act(x, wp, conv._vfa_bnpad, conv.stride, conv.padding,
+                       act)
+        return y[:, :kout].contiguous(memory_format=torch.channels_last)
     if eligible and c % 8 != 0:
         # stems (C=3/2/1) and the RAFT corr input (C=324): zero-pad the
         # channel dim — weight padded once and cached on the module, input
