@@ -205,3 +205,25 @@ def test_resume_skip_does_not_mix_prefetched_decodes(tmp_path):
         got = np.load(feat_dir / f'v{i}_CLIP-ViT-B_32.npy')
         np.testing.assert_allclose(got, solo['CLIP-ViT-B/32'],
                                    rtol=1e-5, atol=1e-6, err_msg=p)
+
+
+def test_clip4clip_auto_checkpoint(tmp_path, monkeypatch, y4m_video):
+    """CLIP4CLIP auto-loads its conventional checkpoint when --weights_path
+    is absent (reference extract_clip.py:58-63 behavior)."""
+    import os
+    import torch
+    from video_features_amd.config import Config
+    from video_features_amd.extractors.clip import ExtractCLIP
+    from video_features_amd.models.clip_vit import VisionTransformer
+    torch.manual_seed(7)
+    m = VisionTransformer()
+    cache = tmp_path / 'wcache'
+    cache.mkdir()
+    torch.save(m.state_dict(), str(cache / 'CLIP4CLIP-ViT-B-32.pth'))
+    monkeypatch.setenv('VFA_WEIGHTS_CACHE', str(cache))
+    cfg = Config(feature_type='CLIP4CLIP-ViT-B-32', video_paths=[y4m_video],
+                 cpu=True, tmp_path=str(tmp_path / 'tmp'))
+    ex = ExtractCLIP(cfg, external_call=True)
+    built = ex.models_for(torch.device('cpu'))
+    for k, v in built.state_dict().items():
+        assert torch.equal(v, m.state_dict()[k]), k

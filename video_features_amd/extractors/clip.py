@@ -32,8 +32,23 @@ class ExtractCLIP(BaseExtractor):
             model = build_clip_resnet(self.feature_type)
         else:
             model = build_clip_vit(self.feature_type)
-        if self.cfg.weights_path:
-            self.load_weights(model, self.cfg.weights_path)
+        wp = self.cfg.weights_path
+        if not wp and self.feature_type == 'CLIP4CLIP-ViT-B-32':
+            # reference convenience: the CLIP4CLIP variant auto-loads its
+            # local checkpoint (reference models/CLIP/extract_clip.py:58-63
+            # uses models/CLIP/checkpoints/CLIP4CLIP-ViT-B-32.pth); here the
+            # conventional spots are ./checkpoints/ and the weights cache
+            from ..utils.weights import cache_dir
+            import os
+            for cand in (os.path.join('checkpoints',
+                                      'CLIP4CLIP-ViT-B-32.pth'),
+                         os.path.join(cache_dir(),
+                                      'CLIP4CLIP-ViT-B-32.pth')):
+                if os.path.exists(cand):
+                    wp = cand
+                    break
+        if wp:
+            self.load_weights(model, wp)
         model = model.to(device=device, dtype=dtype).eval()
         return model
 
