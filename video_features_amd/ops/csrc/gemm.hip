@@ -355,20 +355,29 @@ void linear8p_kernel(const __bf16* __restrict__ a,
           afr[mi2][kk] = *reinterpret_cast<const bf16x8*>(
               la_t + q * 8192 + swz(arow * 128 + kk * 64 + hi4 * 16));
         }
-      // stage: A(T+1, q) and B(T+2, q) — 2 glds, uniform every phase
-      stage_a(T + 1, q);
-      stage_b(T + 2, q);
-      __builtin_amdgcn_s_waitcnt(0x3f76);    // vmcnt(6): 3 phases in flight
+      // vmcnt(4): stages issue MID-phase (between the MFMA halves below),
+      // so leaving 2 phases' stages (4 loads) in flight guarantees the
+      // 3-phase-old stage — and A's stage-to-use distance is 4
+      __builtin_amdgcn_s_waitcnt(0x3f74);
       __builtin_amdgcn_s_barrier();
       __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-      for (int kk = 0; kk < 2; ++kk)
+      for (int mi2 = 0; mi2 < 2; ++mi2)
 #pragma unroll
-        for (int mi2 = 0; mi2 < 2; ++mi2)
+        for (int j = 0; j < 4; ++j)
+          acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mi2][0], bfr[j][0], acc[q * 2 + mi2][j], 0, 0, 0);
+      // the fine interleave: issue this phase's 2 glds between the MFMA
+      // halves so the memory pipe overlaps the math (§5.5: the per-phase
+      // interleave is the lever; a coarse phase-split HURTS)
+      stage_a(T + 1, q);
+      stage_b(T + 2, q);
 #pragma unroll
-          for (int j = 0; j < 4; ++j)
-            acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                afr[mi2][kk], bfr[j][kk], acc[q * 2 + mi2][j], 0, 0, 0);
+      for (int mi2 = 0; mi2 < 2; ++mi2)
+#pragma unroll
+        for (int j = 0; j < 4; ++j)
+          acc[q * 2 + mi2][j] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mi2][1], bfr[j][1], acc[q * 2 + mi2][j], 0, 0, 0);
       __builtin_amdgcn_s_setprio(0);
       __builtin_amdgcn_s_barrier();          // phase lockstep
     }
