@@ -59,10 +59,17 @@ class Conv2Plus1D(nn.Module):
         y = ops.conv2d_act(xf, sw, self.spatial.bias, ss, 1)
         y = _flat_bn_relu(y, self.bn, True)
         w = self.temporal.weight                   # (O, M, 3, 1, 1)
-        o = w.shape[0]
+        o, mid = w.shape[0], w.shape[1]
+        # R(2+1)D mid-planes (144/230/460/921...) are often not %8; on GPU
+        # the weight is zero-padded to c8 once (cached) and the conv
+        # kernel's pad pass widens the input — in-tree instead of MIOpen
+        c8 = ((mid + 7) // 8 * 8
+              if w.is_cuda and w.dtype == torch.bfloat16 else mid)
         wcat = cached_cl_weight(
             self, 'wcat', w,
-            lambda: w.permute(2, 0, 1, 3, 4).reshape(3 * o, w.shape[1], 1, 1))
+            lambda: F.pad(
+                w.permute(2, 0, 1, 3, 4).reshape(3 * o, mid, 1, 1),
+                (0, 0, 0, 0, 0, c8 - mid)))
         bias = self.temporal.bias
         if bias is not None:
             def mk_bcat():
@@ -72,7 +79,7 @@ class Conv2Plus1D(nn.Module):
             bcat = cached_cl_weight(self, 'bcat', bias, mk_bcat)
         else:
             bcat = None
-        y = F.conv2d(y, wcat, bcat)
+        y = ops.conv2d_act(y, wcat, bcat)
         return temporal_merge(y, b, kt=3, st=self.temporal.stride[0], p0=1,
                               relu=relu_after)
 
@@ -111,8 +118,8 @@ class R21DBlock(nn.Module):
             identity = temporal_select(xf, b, st)
             dw = cached_cl_weight(self, 'dw', conv.weight,
                                   lambda: conv.weight[:, :, 0])
-            identity = F.conv2d(identity, dw, conv.bias,
-                                stride=conv.stride[1])
+            identity = ops.conv2d_act(identity, dw, conv.bias,
+                                      conv.stride[1], 0)
             identity = _flat_bn_relu(identity, bn, False)
         bn1_folded = isinstance(self.bn1, nn.Identity)
         out = self.conv1.forward_flat(xf, b, relu_after=bn1_folded)
@@ -160,15 +167,22 @@ class R2Plus1D18(nn.Module):
         # stem: (1,7,7)/s(1,2,2) conv2d -> BN+ReLU -> 3-tap temporal 1x1
         sc0, sbn0, sc1, sbn1 = (self.stem[0], self.stem[1], self.stem[3],
                                 self.stem[4])
-        s0w = cached_cl_weight(self, 's0w', sc0.weight,
-                               lambda: sc0.weight[:, :, 0])
+        stem_c8 = (8 if sc0.weight.is_cuda
+                   and sc0.weight.dtype == torch.bfloat16 else 3)
+        s0w = cached_cl_weight(
+            self, 's0w', sc0.weight,
+            lambda: F.pad(sc0.weight[:, :, 0], (0, 0, 0, 0, 0, stem_c8 - 3)))
         xf = ops.conv2d_act(xf, s0w, sc0.bias, 2, 3)
         xf = _flat_bn_relu(xf, sbn0, True)
-        o = sc1.weight.shape[0]
+        o, mid = sc1.weight.shape[0], sc1.weight.shape[1]
+        c8 = ((mid + 7) // 8 * 8
+              if sc1.weight.is_cuda and sc1.weight.dtype == torch.bfloat16
+              else mid)
         wcat = cached_cl_weight(
             self, 's1w', sc1.weight,
-            lambda: sc1.weight.permute(2, 0, 1, 3, 4).reshape(
-                3 * o, sc1.weight.shape[1], 1, 1))
+            lambda: F.pad(
+                sc1.weight.permute(2, 0, 1, 3, 4).reshape(3 * o, mid, 1, 1),
+                (0, 0, 0, 0, 0, c8 - mid)))
         if sc1.bias is not None:
             def mk_bcat():
                 bc = torch.zeros(3 * o, device=sc1.bias.device,
@@ -179,8 +193,8 @@ class R2Plus1D18(nn.Module):
         else:
             bcat = None
         sbn1_folded = isinstance(sbn1, nn.Identity)
-        xf = temporal_merge(F.conv2d(xf, wcat, bcat), b, kt=3, st=1, p0=1,
-                            relu=sbn1_folded)
+        xf = temporal_merge(ops.conv2d_act(xf, wcat, bcat), b, kt=3, st=1,
+                            p0=1, relu=sbn1_folded)
         if not sbn1_folded:
             xf = _flat_bn_relu(xf, sbn1, True)
         for layer in (self.layer1, self.layer2, self.layer3, self.layer4):
