@@ -450,6 +450,10 @@ def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
             if res is not None else None
         return _ext.conv2d_nhwc(x, w_cl, bias, r, sh, sw, pt, pb, pl, pr,
                                 _ACT_IDS[act])
+    if weight.shape[1] > x.shape[1]:
+        # caller passed a channel-padded weight but the GPU path declined
+        # (VFA_NO_CONV, forced torch ops, ...): slice the zero pad back off
+        weight = weight[:, :x.shape[1]]
     if pt != pb or pl != pr:
         x = torch.nn.functional.pad(x, (pl, pr, pt, pb))
         pt = pl = 0
