@@ -221,19 +221,40 @@ class BasicMotionEncoder(nn.Module):
         self.convf2 = nn.Conv2d(128, 64, 3, 1, 1)
         self.conv = nn.Conv2d(192 + 64, 128 - 2, 3, 1, 1)
 
-    def forward(self, flow, corr):
-        c = ops.conv2d_mod(self.convc2,
-                           ops.conv2d_mod(self.convc1, corr, 'relu'),
-                           'relu')
-        f = ops.conv2d_mod(self.convf2,
-                           ops.conv2d_mod(self.convf1, flow, 'relu'),
-                           'relu')
+    def forward(self, flow, corr, out: torch.Tensor = None,
+                out_off: int = 0):
+        c1 = ops.conv2d_mod(self.convc1, corr, 'relu')
+        f1 = ops.conv2d_mod(self.convf1, flow, 'relu')
+        if (out is not None and c1.is_cuda and c1.dtype == torch.bfloat16
+                and ops.hip_available()
+                and c1.is_contiguous(memory_format=torch.channels_last)):
+            # cat-free path: convc2/convf2 write their channel slices of
+            # ONE buffer via the conv epilogue's ldc, and the final conv
+            # writes straight into the caller's (GRU input) buffer — the
+            # three CatArrayBatchedCopy passes disappear
+            b, _, h, w = c1.shape
+            cat = torch.empty((b, h, w, 256), device=c1.device,
+                              dtype=c1.dtype).permute(0, 3, 1, 2)
+            ops.conv2d_act(c1, self.convc2.weight, self.convc2.bias, 1, 1,
+                           'relu', out=cat, out_off=0)
+            ops.conv2d_act(f1, self.convf2.weight, self.convf2.bias, 1, 1,
+                           'relu', out=cat, out_off=192)
+            ops.conv2d_act(cat, self.conv.weight, self.conv.bias, 1, 1,
+                           'relu', out=out, out_off=out_off)
+            out[:, out_off + 126:out_off + 128] = flow
+            return None
+        c = ops.conv2d_mod(self.convc2, c1, 'relu')
+        f = ops.conv2d_mod(self.convf2, f1, 'relu')
         cat = torch.cat([c, f], dim=1)
         if cat.is_cuda and not cat.is_contiguous(
                 memory_format=torch.channels_last):
             cat = cat.contiguous(memory_format=torch.channels_last)
-        out = ops.conv2d_mod(self.conv, cat, 'relu')
-        return torch.cat([out, flow], dim=1)
+        res = torch.cat([ops.conv2d_mod(self.conv, cat, 'relu'), flow],
+                        dim=1)
+        if out is not None:
+            out[:, out_off:out_off + 128] = res
+            return None
+        return res
 
 
 class BasicUpdateBlock(nn.Module):
@@ -313,9 +334,10 @@ class RAFT(nn.Module):
         for it in range(iters):
             corr = corr_fn(coords1, nhwc=self.nhwc, out_dtype=dtype)
             flow = (coords1 - coords0).to(dtype).contiguous(memory_format=mf)
-            motion = ub.encoder(flow, corr)
-            hx[:, ctx_end:] = motion
-            rhx[:, ctx_end:] = motion
+            # motion features written straight into hx's motion slice by
+            # the encoder's conv epilogue; rhx mirrors with one copy
+            ub.encoder(flow, corr, out=hx, out_off=ctx_end)
+            rhx[:, ctx_end:] = hx[:, ctx_end:]
             ub.gru(hx, rhx, self.nhwc)
             net = hx[:, :self.hdim]
             if self.nhwc:

@@ -420,7 +420,9 @@ def conv1x1_act(x: torch.Tensor, weight: torch.Tensor,
 def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
                bias: Optional[torch.Tensor] = None,
                stride=1, padding=0, act: str = 'none',
-               res: Optional[torch.Tensor] = None) -> torch.Tensor:
+               res: Optional[torch.Tensor] = None,
+               out: Optional[torch.Tensor] = None,
+               out_off: int = 0) -> torch.Tensor:
     """Fused conv2d + bias + activation (+ residual) on channels_last bf16.
 
     GPU path: the in-tree implicit-GEMM MFMA kernel (conv2d.hip) — the
@@ -449,7 +451,7 @@ def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
         r = res.contiguous(memory_format=torch.channels_last) \
             if res is not None else None
         return _ext.conv2d_nhwc(x, w_cl, bias, r, sh, sw, pt, pb, pl, pr,
-                                _ACT_IDS[act])
+                                _ACT_IDS[act], out, out_off)
     if weight.shape[1] > x.shape[1]:
         # caller passed a channel-padded weight but the GPU path declined
         # (VFA_NO_CONV, forced torch ops, ...): slice the zero pad back off
@@ -470,6 +472,9 @@ def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
         y = torch.nn.functional.gelu(y, approximate='tanh')
     elif act == 'leaky_relu':
         y = torch.nn.functional.leaky_relu(y, 0.1)
+    if out is not None:
+        out[:, out_off:out_off + y.shape[1]] = y
+        return out
     return y
 
 

@@ -6,6 +6,7 @@
 #include <mutex>
 #include <unordered_map>
 #include <hip/hip_runtime.h>
+#include <hip/hip_bf16.h>
 
 extern "C" {
 void vfa_quick_gelu(const void*, void*, long long, int, hipStream_t);
@@ -52,7 +53,7 @@ void vfa_temporal_merge(const void*, void*, int, int, int, int, int, int,
                         int, long long, int, int, hipStream_t);
 void vfa_conv2d_nhwc(const void*, const void*, const void*, const void*,
                      const void*, void*, int, int, int, int, int, int, int,
-                     int, int, int, int, int, int, int, hipStream_t);
+                     int, int, int, int, int, int, int, int, hipStream_t);
 void vfa_pad2d_nhwc(const void*, void*, int, int, int, int, int, int, int,
                     int, int, hipStream_t);
 }
@@ -424,7 +425,9 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
                           c10::optional<torch::Tensor> res,
                           int64_t stride_h, int64_t stride_w,
                           int64_t pt, int64_t pb, int64_t pl, int64_t pr,
-                          int64_t act) {
+                          int64_t act,
+                          c10::optional<torch::Tensor> out_buf,
+                          int64_t out_off) {
   // x (B, C, H, W) channels_last bf16; w (K, C, KH, KW) channels_last bf16
   // (physical (K, KH, KW, C) = the (N, Kr) B-operand); out (B, K, OH, OW)
   // channels_last.  Implicit-GEMM MFMA kernel; input pre-padded by the
@@ -486,18 +489,35 @@ torch::Tensor conv2d_nhwc(torch::Tensor x, torch::Tensor w,
                 "residual must be CL bf16 of the output shape");
     rptr = res->data_ptr();
   }
-  auto out = torch::empty({(long)b, oh, ow, kout}, x.options())
-                 .permute({0, 3, 1, 2});
+  // optional preallocated WIDER output buffer: the epilogue writes the
+  // channel slice [out_off, out_off + kout) of a (B, C_total, OH, OW) CL
+  // tensor (eliminates a following cat copy)
+  torch::Tensor out;
+  int ldc = kout;
+  void* optr;
+  if (out_buf.has_value()) {
+    out = *out_buf;
+    TORCH_CHECK(cl_contig(out) && out.scalar_type() == torch::kBFloat16 &&
+                out.size(0) == b && out.size(2) == oh && out.size(3) == ow &&
+                out_off + kout <= out.size(1),
+                "out buffer must be CL bf16 (B, >=off+kout, OH, OW)");
+    ldc = (int)out.size(1);
+    optr = static_cast<char*>(out.data_ptr()) + out_off * 2;
+  } else {
+    out = torch::empty({(long)b, oh, ow, kout}, x.options())
+              .permute({0, 3, 1, 2});
+    optr = out.data_ptr();
+  }
   if (inline_pad)
     vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr,
-                    zpage.data_ptr(), out.data_ptr(), b, h, ww, c8, oh, ow,
+                    zpage.data_ptr(), optr, b, h, ww, c8, oh, ow,
                     kout, kh, kw, (int)stride_h, (int)stride_w, (int)pt,
-                    (int)pl, (int)act, stream);
+                    (int)pl, ldc, (int)act, stream);
   else
     vfa_conv2d_nhwc(xp.data_ptr(), w.data_ptr(), bptr, rptr,
-                    zpage.data_ptr(), out.data_ptr(), b, hp, wp, c8, oh,
+                    zpage.data_ptr(), optr, b, hp, wp, c8, oh,
                     ow, kout, kh, kw, (int)stride_h, (int)stride_w, 0, 0,
-                    (int)act, stream);
+                    ldc, (int)act, stream);
   return out;
 }
 

@@ -182,7 +182,7 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
                         const __bf16* __restrict__ res,
                         const __bf16* __restrict__ zp,
                         __bf16* __restrict__ out, ConvGeom g, int mtotal,
-                        int tiles_m, int tiles_n) {
+                        int ldc, int tiles_m, int tiles_n) {
   constexpr int MI = BM / (WAVES / WN) / 16;   // 16-row MFMA tiles / wave
   constexpr int NJ = BN / WN / 16;             // 16-col MFMA tiles / wave
   static_assert(NJ == 4, "epilogue assumes 4 column fragments per wave");
@@ -301,7 +301,9 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
         if (!tile_full && row >= mtotal) continue;
         float v = acc[i][j][r] + bv;
         if (res) v += (float)res[(long long)row * n + col];
-        out[(long long)row * n + col] = (__bf16)conv_act_f(v, ACT);
+        // ldc > n: writing a channel slice of a wider NHWC buffer (the
+        // caller eliminated a cat copy); res stays logically (M, n)
+        out[(long long)row * ldc + col] = (__bf16)conv_act_f(v, ACT);
       }
     }
   }
@@ -310,7 +312,7 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
 template <int ACT, int BM, int BN, int WAVES, int WN>
 void launch_cfg(const void* x, const void* w, const void* bias,
                 const void* res, const void* zp, void* out,
-                const ConvGeom& g, int m, hipStream_t stream) {
+                const ConvGeom& g, int m, int ldc, hipStream_t stream) {
   const int tiles_m = (m + BM - 1) / BM, tiles_n = (g.kout + BN - 1) / BN;
   const dim3 grid(tiles_m * tiles_n);
   const size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
@@ -318,13 +320,13 @@ void launch_cfg(const void* x, const void* w, const void* bias,
                      dim3(WAVES * 64), lds, stream, (const __bf16*)x,
                      (const __bf16*)w, (const __bf16*)bias,
                      (const __bf16*)res, (const __bf16*)zp, (__bf16*)out,
-                     g, m, tiles_m, tiles_n);
+                     g, m, ldc, tiles_m, tiles_n);
 }
 
 template <int ACT>
 void launch_conv(const void* x, const void* w, const void* bias,
                  const void* res, const void* zp, void* out,
-                 const ConvGeom& g, hipStream_t stream) {
+                 const ConvGeom& g, int ldc, hipStream_t stream) {
   const int m = g.b * g.oh * g.ow;
   // pick the tile minimizing padded work, preferring the bigger tile when
   // waste ties and the grid still fills the chip (256 CUs)
@@ -339,11 +341,11 @@ void launch_conv(const void* x, const void* w, const void* bias,
   const long long c128 = cost(128, 128, 120);
   const long long c64 = cost(256, 64, 120);
   if (c256 <= c128 && c256 <= c64)
-    launch_cfg<ACT, 256, 256, 8, 4>(x, w, bias, res, zp, out, g, m, stream);
+    launch_cfg<ACT, 256, 256, 8, 4>(x, w, bias, res, zp, out, g, m, ldc, stream);
   else if (c64 < c128)
-    launch_cfg<ACT, 256, 64, 4, 1>(x, w, bias, res, zp, out, g, m, stream);
+    launch_cfg<ACT, 256, 64, 4, 1>(x, w, bias, res, zp, out, g, m, ldc, stream);
   else
-    launch_cfg<ACT, 128, 128, 4, 2>(x, w, bias, res, zp, out, g, m, stream);
+    launch_cfg<ACT, 128, 128, 4, 2>(x, w, bias, res, zp, out, g, m, ldc, stream);
 }
 
 // ------------------------------------------------------------- pad kernel
@@ -391,19 +393,21 @@ extern "C" {
 // act: 0 none, 1 relu, 2 quick_gelu, 3 gelu_tanh, 4 leaky_relu(0.1)
 // pt/pl: inline zero-pad applied by the kernel (zp = 1 KiB zero page);
 // pass 0 with pre-padded h/w when the pad was materialized
+// ldc: output row stride in elements (= kout normally; larger when the
+// epilogue writes a channel slice of a wider NHWC buffer)
 void vfa_conv2d_nhwc(const void* x, const void* w, const void* bias,
                      const void* res, const void* zp, void* out, int b,
                      int hp, int wp, int cin, int oh, int ow, int kout,
                      int kh, int kw, int sh, int sw, int pt, int pl,
-                     int act, hipStream_t stream) {
+                     int ldc, int act, hipStream_t stream) {
   ConvGeom g{b, hp, wp, cin, oh, ow, kout, kh, kw, sh, sw, pt, pl,
              kh * kw * cin};
   switch (act) {
-    case 0: launch_conv<0>(x, w, bias, res, zp, out, g, stream); break;
-    case 1: launch_conv<1>(x, w, bias, res, zp, out, g, stream); break;
-    case 2: launch_conv<2>(x, w, bias, res, zp, out, g, stream); break;
-    case 3: launch_conv<3>(x, w, bias, res, zp, out, g, stream); break;
-    case 4: launch_conv<4>(x, w, bias, res, zp, out, g, stream); break;
+    case 0: launch_conv<0>(x, w, bias, res, zp, out, g, ldc, stream); break;
+    case 1: launch_conv<1>(x, w, bias, res, zp, out, g, ldc, stream); break;
+    case 2: launch_conv<2>(x, w, bias, res, zp, out, g, ldc, stream); break;
+    case 3: launch_conv<3>(x, w, bias, res, zp, out, g, ldc, stream); break;
+    case 4: launch_conv<4>(x, w, bias, res, zp, out, g, ldc, stream); break;
   }
 }
 
