@@ -485,8 +485,11 @@ def main():
     p.add_argument('--warmup', type=int, default=4)
     p.add_argument('--model', choices=['clip', 'i3d_raft', 'resnet50', 'vggish_r21d'],
                    default='clip')
-    p.add_argument('--videos-per-step', type=int, default=64,
-                   help='CLIP: synthetic videos (x12 frames) per rank per step')
+    p.add_argument('--videos-per-step', type=int, default=256,
+                   help='CLIP: synthetic videos (x12 frames) per rank per '
+                        'step (256 = the measured throughput knee; per-step '
+                        'launch/copy overheads amortize: 65.7k -> 71.4k f/s '
+                        'vs 64)')
     p.add_argument('--frame-batch', type=int, default=384,
                    help='CLIP: frames per forward chunk')
     p.add_argument('--clips-per-step', type=int, default=None,
