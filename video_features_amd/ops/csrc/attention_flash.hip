@@ -103,17 +103,19 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
   const __bf16* q_g = qkv + ((long long)bi * n + qbase) * row_stride +
                       (long long)hi * D;                    // +0 for q
   const int n_kv0 = (n + KVBLK - 1) / KVBLK;
-  stage_rows(q_g, QBLK, row_stride, s_q, max(0, n - qbase));
   if (n_kv0 == 1) {
-    // ViT-length sequences (N <= 64): one KV tile — stage Q, K, V together
-    // so all staging loads are in flight at once behind a SINGLE barrier
-    // (PMC: this kernel was 66% wave-parked with the 3-barrier schedule)
+    // ViT-length sequences (N <= 64): one KV tile — stage K and V behind a
+    // SINGLE barrier (PMC: this kernel was 66% wave-parked with the
+    // 3-barrier schedule); Q skips LDS entirely — each Q row is read by
+    // exactly ONE wave, so its fragments load straight from global below
     const __bf16* k_g0 = qkv + (long long)bi * n * row_stride +
                          ((long long)1 * h_total + hi) * D;
     const __bf16* v_g0 = qkv + (long long)bi * n * row_stride +
                          ((long long)2 * h_total + hi) * D;
     stage_rows(k_g0, KVBLK, row_stride, s_k, n);
     stage_vt(v_g0, row_stride, s_vt, n);
+  } else {
+    stage_rows(q_g, QBLK, row_stride, s_q, max(0, n - qbase));
   }
   __syncthreads();
 
@@ -144,10 +146,19 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
     f32x4 s_frag[4];
 #pragma unroll
     for (int t = 0; t < 4; ++t) s_frag[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+    // Q A-fragment: straight from global in the single-tile case (valid
+    // rows clamped — rows >= n are never written in the epilogue)
+    const int qv = max(0, n - qbase);
+    const int qr = min(wave * 16 + lo, max(qv - 1, 0));
 #pragma unroll
     for (int kt = 0; kt < 2; ++kt) {
-      bf16x8 a = *reinterpret_cast<const bf16x8*>(
-          s_q + (wave * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
+      bf16x8 a;
+      if (n_kv == 1)
+        a = *reinterpret_cast<const bf16x8*>(
+            q_g + (long long)qr * row_stride + kt * 32 + hi4 * 8);
+      else
+        a = *reinterpret_cast<const bf16x8*>(
+            s_q + (wave * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
 #pragma unroll
       for (int nt = 0; nt < 4; ++nt) {
         bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
