@@ -33,8 +33,10 @@ namespace {
 
 constexpr int BK = 64;
 
+// see gemm.hip swz(): conflict-free row-slot XOR ((row>>1)&7, searched
+// against gfx950's REAL mixed b128 lane groups) for the 128-B-row images
 __device__ __forceinline__ int cswz(int byte_off) {
-  return byte_off ^ (((byte_off >> 9) & 1) << 5);
+  return byte_off ^ (((byte_off >> 8) & 7) << 4);
 }
 
 __device__ __forceinline__ float conv_act_f(float x, int kind) {
@@ -83,18 +85,21 @@ template <int ROWS, int WAVES>
 __device__ __forceinline__ void conv_stage_a(
     const __bf16* __restrict__ x, const __bf16* __restrict__ zp,
     const ConvGeom& g, const PixRef* __restrict__ aref, int k0,
-    char* lds_base, int wave, int lane, int kfrac, int piece, int npieces) {
+    char* lds_base, int wave, int lane, int kfrac0, int arin_half,
+    int piece, int npieces) {
   constexpr int NSUB = ROWS * 128 / 1024;
   constexpr int PER_WAVE = NSUB / WAVES;
-  const int k = k0 + kfrac;
-  const int tap = k / g.cin;            // uniform-cost u32 div
-  const int c = k - tap * g.cin;
-  const int r = tap / g.kw;
-  const int s = tap - r * g.kw;
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
     if (piece >= 0 && (i * npieces) / PER_WAVE != piece) continue;
     const int sub = wave * PER_WAVE + i;
+    // per-subtile swizzle: f = ((sub&1)<<2)|(r_in>>1) on element bits 3..5
+    const int k = k0 + (kfrac0 ^
+                        ((((sub & 1) << 2) | arin_half) << 3));
+    const int tap = k / g.cin;          // uniform-cost u32 div
+    const int c = k - tap * g.cin;
+    const int r = tap / g.kw;
+    const int s = tap - r * g.kw;
     const int iy = aref[i].y0 + r, ix = aref[i].x0 + s;
     const bool ok = (unsigned)iy < (unsigned)g.hp &&
                     (unsigned)ix < (unsigned)g.wp;
@@ -114,14 +119,15 @@ __device__ __forceinline__ void conv_stage_w(
     int wave, int lane, int piece, int npieces) {
   constexpr int NSUB = ROWS * 128 / 1024;
   constexpr int PER_WAVE = NSUB / WAVES;
+  // subtile-local swizzle; f = ((sub&1)<<2)|(r_in>>1) (see gemm.hip)
   const int off = lane * 16;
-  const int off_log = off ^ (((off >> 9) & 1) << 5);
-  const int r_in = off_log >> 7;
-  const int b_in = off_log & 127;
+  const int r_in = off >> 7;
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
     if (piece >= 0 && (i * npieces) / PER_WAVE != piece) continue;
     const int sub = wave * PER_WAVE + i;
+    const int b_in = (off & 127) ^
+                     ((((sub & 1) << 2) | (r_in >> 1)) << 4);
     const __bf16* src = wgt + (long long)(sub * 8 + r_in) * row_stride;
     __builtin_amdgcn_global_load_lds(
         reinterpret_cast<const unsigned int*>(
@@ -216,9 +222,8 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
 
   // per-thread A row refs for the glds path (same rows every K-tile);
   // the per-lane swizzled byte offset contributes kfrac elements
-  const int off_log = (lane * 16) ^ ((((lane * 16) >> 9) & 1) << 5);
-  const int a_rin = off_log >> 7;
-  const int kfrac = (off_log & 127) >> 1;
+  const int a_rin = (lane * 16) >> 7;
+  const int kfrac0 = ((lane * 16) & 127) >> 1;   // pre-swizzle elements
   PixRef aref[PER_WAVE];
 #pragma unroll
   for (int i = 0; i < PER_WAVE; ++i) {
@@ -231,7 +236,7 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
     const int k0 = kt * BK;
     if (tile_full && (kfull || kt + 1 < nk)) {
       conv_stage_a<BM, WAVES>(x, zp, g, aref, k0, sA(buf), wave, lane,
-                              kfrac, -1, 1);
+                              kfrac0, a_rin >> 1, -1, 1);
       conv_stage_w<BN, WAVES>(wgt + (long long)n0 * kr + k0, kr, sB(buf),
                               wave, lane, -1, 1);
     } else {
@@ -273,7 +278,7 @@ void conv2d_nhwc_kernel(const __bf16* __restrict__ x,
       }
       if (glds_nxt) {
         conv_stage_a<BM, WAVES>(x, zp, g, aref, k0_nxt, sA(1 - cur), wave,
-                                lane, kfrac, kk, 2);
+                                lane, kfrac0, a_rin >> 1, kk, 2);
         conv_stage_w<BN, WAVES>(wgt + (long long)n0 * kr + k0_nxt, kr,
                                 sB(1 - cur), wave, lane, kk, 2);
       }
