@@ -655,12 +655,15 @@ void launch_linear(const void* a, const void* w, const void* bias,
   // tail tile and the block-round quantization cost more than the smaller
   // tile's overhead — measured 339 vs 528 TF at M=9600) and the grid
   // still fills the chip
-  // 8-phase pipelined kernel: correct but currently BELOW the 2-buffer
-  // structure (1018 vs 1149 TF @8k — the fine per-phase interleave is
-  // still missing, cdna guide §5.5); opt-in via VFA_8P for experiments
-  static const bool use8p = getenv("VFA_8P") != nullptr;
+  // 8-phase pipelined kernel: with the conflict-free LDS swizzle it wins
+  // the act-none shapes (qkv 742 vs 714 TF, 8192^3 1160 vs 1104) but
+  // trails the 2-buffer kernel's fused-activation epilogue on fc1 — route
+  // by activation.  VFA_NO_8P / VFA_8P force either way for A/B.
+  static const int env8p = getenv("VFA_8P") ? 1
+                           : getenv("VFA_NO_8P") ? -1 : 0;
   if (launch_thin<ACT>(a, w, bias, res, c, m, n, k, stream)) return;
-  if (use8p && launch_8p<ACT>(a, w, bias, res, c, m, n, k, stream)) return;
+  const bool want8p = env8p > 0 || (env8p == 0 && ACT == 0);
+  if (want8p && launch_8p<ACT>(a, w, bias, res, c, m, n, k, stream)) return;
   const long long tiles = (long long)((m + 255) / 256) * ((n + 255) / 256);
   if (n >= 256 && m % 256 == 0 && tiles >= 150)
     launch_tile<ACT, true>(a, w, bias, res, c, m, n, k, stream);
