@@ -57,6 +57,9 @@ def test_linear_act(dev, m, n, k, act):
     (131072, 128, 256),   # l2 conv1
     (70000, 512, 128),    # l2 conv3 (ragged M)
     (65552, 48, 96),      # ragged N + M just over the threshold
+    (131072, 192, 144),   # r21d temporal conv regime: K%32 != 0 (masked
+                          # tail fragment + zero-filled W rows)
+    (131072, 256, 232),   # K%32 = 8
 ])
 @pytest.mark.parametrize('act', ['none', 'relu'])
 def test_linear_thin_streaming(dev, m, n, k, act):

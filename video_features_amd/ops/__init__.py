@@ -447,6 +447,15 @@ def conv2d_act(x: torch.Tensor, weight: torch.Tensor,
             and weight.shape[1] == (x.shape[1] + 7) // 8 * 8
             and x.is_contiguous(memory_format=torch.channels_last)
             and not _env_flag('VFA_NO_CONV')):
+        if (weight.shape[2] == 1 and weight.shape[3] == 1
+                and (sh, sw) == (1, 1) and (pt, pb, pl, pr) == (0, 0, 0, 0)
+                and weight.shape[1] == x.shape[1] and out is None
+                and weight.shape[0] % 8 == 0):
+            # pure 1x1: the linear stack's router picks the right GEMM
+            # structure — in particular the thin-K streaming kernel for
+            # the M-huge R21D temporal convs (profiled 3.5x off the memory
+            # floor on the tiled conv path at their ragged K)
+            return conv1x1_act(x, weight, bias, act, res)
         w_cl = weight.contiguous(memory_format=torch.channels_last)
         r = res.contiguous(memory_format=torch.channels_last) \
             if res is not None else None

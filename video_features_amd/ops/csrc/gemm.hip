@@ -467,29 +467,36 @@ void linear_thin_kernel(const __bf16* __restrict__ a,
                         const __bf16* __restrict__ w,
                         const __bf16* __restrict__ bias,
                         const __bf16* __restrict__ res,
-                        __bf16* __restrict__ c, int m, int n, int nch) {
+                        __bf16* __restrict__ c, int m, int n, int nch,
+                        int k) {
   // grid: (m_blocks, n_outer); block 256 = 4 waves, each wave 32 rows.
-  // KF = K/32 is COMPILE-TIME: runtime-indexed ext_vector arrays go to
-  // scratch (cdna_hip_programming.md §5.4 rule 20 — the first version of
-  // this kernel hit exactly that: 4 TF/s from scratch traffic)
-  constexpr int K = KF * 32;
+  // KF = ceil(K/32) is COMPILE-TIME: runtime-indexed ext_vector arrays go
+  // to scratch (cdna_hip_programming.md §5.4 rule 20 — the first version
+  // of this kernel hit exactly that: 4 TF/s from scratch traffic).
+  // K itself may be any multiple of 8 (the R21D temporal 1x1 convs have
+  // K = mid-planes like 144/232): each lane's last A fragment is either
+  // fully valid or fully masked to zero (8-elem granularity), and the W
+  // LDS rows are zero-filled up to the KC boundary.
+  constexpr int KC = KF * 32;
   const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   const int lo = lane & 15, hi4 = lane >> 4;
 
-  // LDS layout: W chunk (rows of K+8 elems, breaks the ds_read bank
-  // pattern) + per-wave 16x64 bf16 epilogue bounce tiles (row pitch 72)
+  // LDS layout: W chunk (rows of KC+8 elems, breaks the ds_read bank
+  // pattern; K..KC zero-filled) + per-wave 16x64 bf16 epilogue tiles
   extern __shared__ __attribute__((aligned(16))) char smem_t[];
   __bf16* wl = reinterpret_cast<__bf16*>(smem_t);
-  constexpr int LDW = K + 8;
+  constexpr int LDW = KC + 8;
   const int n0 = blockIdx.y * nch;
   const int ncols = min(nch, n - n0);
   __bf16* et = reinterpret_cast<__bf16*>(smem_t) + (long long)nch * LDW +
                wave * 16 * 72;
-  for (int t = threadIdx.x; t < ncols * (K / 8); t += 256) {
-    const int row = t / (K / 8), seg = t % (K / 8);
-    *reinterpret_cast<uint4*>(wl + row * LDW + seg * 8) =
-        *reinterpret_cast<const uint4*>(w + (long long)(n0 + row) * K +
-                                        seg * 8);
+  for (int t = threadIdx.x; t < ncols * (KC / 8); t += 256) {
+    const int row = t / (KC / 8), seg = t % (KC / 8);
+    uint4 v = {0u, 0u, 0u, 0u};
+    if (seg * 8 < k)
+      v = *reinterpret_cast<const uint4*>(w + (long long)(n0 + row) * k +
+                                          seg * 8);
+    *reinterpret_cast<uint4*>(wl + row * LDW + seg * 8) = v;
   }
   __syncthreads();
 
@@ -502,10 +509,14 @@ void linear_thin_kernel(const __bf16* __restrict__ a,
     for (int mi = 0; mi < 2; ++mi) {
       long long row = r0 + mi * 16 + lo;
       if (row >= m) row = m - 1;               // clamped load, masked store
-      const __bf16* ap = a + row * K + hi4 * 8;
+      const __bf16* ap = a + row * k + hi4 * 8;
 #pragma unroll
-      for (int kk = 0; kk < KF; ++kk)
-        afr[mi][kk] = *reinterpret_cast<const bf16x8*>(ap + kk * 32);
+      for (int kk = 0; kk < KF; ++kk) {
+        if (kk * 32 + hi4 * 8 < k)
+          afr[mi][kk] = *reinterpret_cast<const bf16x8*>(ap + kk * 32);
+        else
+          afr[mi][kk] = bf16x8{};              // masked K tail (x * 0-W)
+      }
     }
     // 64 columns per outer step: the epilogue then sees 128-B contiguous
     // output rows (the naive fragment-layout epilogue's 2-B scalar
@@ -587,13 +598,13 @@ void linear_thin_kernel(const __bf16* __restrict__ a,
 
 template <int ACT, int KF>
 void launch_thin_kf(const void* a, const void* w, const void* bias,
-                    const void* res, void* c, int m, int n, int nch,
+                    const void* res, void* c, int m, int n, int nch, int k,
                     int m_blocks, size_t lds, hipStream_t stream) {
   hipLaunchKernelGGL((linear_thin_kernel<ACT, KF>),
                      dim3(m_blocks, (n + nch - 1) / nch), dim3(256), lds,
                      stream, (const __bf16*)a, (const __bf16*)w,
                      (const __bf16*)bias, (const __bf16*)res, (__bf16*)c,
-                     m, n, nch);
+                     m, n, nch, k);
 }
 
 template <int ACT>
@@ -601,24 +612,27 @@ bool launch_thin(const void* a, const void* w, const void* bias,
                  const void* res, void* c, int m, int n, int k,
                  hipStream_t stream) {
   // n % 8: the vectorized epilogue's 16-B res/out accesses need 8-elem
-  // row alignment
-  if (k > 256 || k % 32 != 0 || m < 65536 || n % 8 != 0) return false;
+  // row alignment.  K any multiple of 8 (<= 256): the last A fragment is
+  // masked per lane, W LDS rows zero-fill to the 32-elem boundary.
+  if (k > 256 || k % 8 != 0 || k < 64 || m < 65536 || n % 8 != 0)
+    return false;
+  const int kc = (k + 31) / 32 * 32;
   // W chunk + 9 KiB epilogue tiles bounded by 80 KiB LDS (2 blocks/CU)
-  const int nch_cap = 36352 / (k + 8) / 16 * 16;
+  const int nch_cap = 36352 / (kc + 8) / 16 * 16;
   const int nch = min((n + 15) & ~15, nch_cap);
   if (nch < 16) return false;
   // enough M-blocks to fill the chip; grid-stride handles the rest
   const int m_blocks = (int)min(((long long)m + 127) / 128, 4096LL);
-  const size_t lds = (size_t)nch * (k + 8) * 2 + 4 * 16 * 72 * 2;
-  switch (k / 32) {
-    case 2: launch_thin_kf<ACT, 2>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    case 3: launch_thin_kf<ACT, 3>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    case 4: launch_thin_kf<ACT, 4>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    case 5: launch_thin_kf<ACT, 5>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    case 6: launch_thin_kf<ACT, 6>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    case 7: launch_thin_kf<ACT, 7>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    case 8: launch_thin_kf<ACT, 8>(a, w, bias, res, c, m, n, nch, m_blocks, lds, stream); break;
-    default: return false;                     // K=32: not worth MFMA
+  const size_t lds = (size_t)nch * (kc + 8) * 2 + 4 * 16 * 72 * 2;
+  switch (kc / 32) {
+    case 2: launch_thin_kf<ACT, 2>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    case 3: launch_thin_kf<ACT, 3>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    case 4: launch_thin_kf<ACT, 4>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    case 5: launch_thin_kf<ACT, 5>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    case 6: launch_thin_kf<ACT, 6>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    case 7: launch_thin_kf<ACT, 7>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    case 8: launch_thin_kf<ACT, 8>(a, w, bias, res, c, m, n, nch, k, m_blocks, lds, stream); break;
+    default: return false;                     // K<64: not worth MFMA
   }
   return true;
 }
