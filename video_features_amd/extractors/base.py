@@ -167,17 +167,31 @@ class BaseExtractor(torch.nn.Module):
             except Exception:
                 pass
         pipeline = type(self).prepare is not None and len(idx_list) > 1
-        pool = fut = None
+        pool = None
+        futures = {}                   # position -> decode future
         if pipeline:
+            import os as _os
             from concurrent.futures import ThreadPoolExecutor
-            pool = ThreadPoolExecutor(max_workers=1,
+            # decode-ahead depth (VFA_DECODE_AHEAD, default 2): the
+            # extractor path is file-decode-bound, so overlapping more
+            # than one video's decode with GPU compute pays; memory grows
+            # by one decoded video per extra worker
+            depth = max(1, int(_os.environ.get('VFA_DECODE_AHEAD', '2')))
+            pool = ThreadPoolExecutor(max_workers=depth,
                                       thread_name_prefix='vfa-decode')
+
+            def fill(from_pos):
+                for p in range(from_pos,
+                               min(from_pos + depth, len(idx_list))):
+                    if p not in futures:
+                        futures[p] = pool.submit(
+                            self.prepare, self.path_list[idx_list[p]])
         for pos, idx in enumerate(idx_list):
             video_path = self.path_list[idx]
-            # a pending future (submitted last iteration) is ALWAYS for
-            # THIS video: take ownership first, so a resume-skip or a
-            # raise can never hand video i's decode to video i+1
-            pending, fut = fut, None
+            # the future at THIS position is always for THIS video: take
+            # ownership first, so a resume-skip or a raise can never hand
+            # video i's decode to video i+1
+            pending = futures.pop(pos, None)
             try:
                 if (self.cfg.resume and not self.external_call
                         and self._already_done(video_path)):
@@ -197,12 +211,10 @@ class BaseExtractor(torch.nn.Module):
                     continue
                 prepared = None
                 if pipeline:
+                    fill(pos + 1)      # keep the decode pipeline full
                     with self._prof('decode'):
                         prepared = (pending.result() if pending is not None
                                     else self.prepare(video_path))
-                    if pos + 1 < len(idx_list):
-                        nxt = self.path_list[idx_list[pos + 1]]
-                        fut = pool.submit(self.prepare, nxt)
                 feats_dict = self.extract(device, models, video_path,
                                           prepared)
                 if self.external_call:
