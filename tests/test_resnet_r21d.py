@@ -116,3 +116,19 @@ def test_r21d_flat_path_matches_5d_reference():
     assert out.shape == expect.shape == (2, 512)
     assert torch.allclose(out, expect, atol=2e-4, rtol=1e-4), \
         (out - expect).abs().max().item()
+
+
+def test_conv2d_act_out_buffer_slice():
+    """conv2d_act writes a channel slice of a wider NHWC buffer (the
+    cat-elimination path); CPU fallback must match the kernel contract."""
+    import torch.nn.functional as F
+    from video_features_amd import ops
+    torch.manual_seed(0)
+    x = torch.randn(2, 16, 8, 8)
+    w = torch.randn(24, 16, 3, 3)
+    buf = torch.zeros(2, 40, 8, 8)
+    out = ops.conv2d_act(x, w, None, 1, 1, 'relu', out=buf, out_off=8)
+    assert out is buf
+    ref = F.relu(F.conv2d(x, w, None, 1, 1))
+    torch.testing.assert_close(buf[:, 8:32], ref)
+    assert (buf[:, :8] == 0).all() and (buf[:, 32:] == 0).all()

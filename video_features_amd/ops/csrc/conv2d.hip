@@ -19,8 +19,9 @@
 //
 // Machinery shared with gemm.hip (cdna_hip_programming.md §5): 256x256 /
 // 128x128 tiles, BK=64, double-buffered LDS staged by
-// __builtin_amdgcn_global_load_lds width 16 with the st_16x32 XOR swizzle
-// on the per-lane *source* address (LDS image stays lane-linear), glds for
+// __builtin_amdgcn_global_load_lds width 16 with the conflict-free XOR
+// swizzle (cswz) on the per-lane *source* address (lane-linear LDS
+// image), glds for
 // the next K-tile spread across the current tile's two MFMA half-steps,
 // mfma_f32_16x16x32_bf16, fused bias+activation(+residual) epilogue.
 // Requires C % 8 == 0 (each lane's 16-B segment stays inside one tap).

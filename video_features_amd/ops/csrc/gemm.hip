@@ -11,10 +11,10 @@
 // operands stage through LDS as [row][64] bf16 images (A rows = m, B rows
 // = n; the B fragment of C[m][n] = dot_k A[m][k] W[n][k] reads the same
 // row-major image as A).  Staging uses __builtin_amdgcn_global_load_lds
-// width 16 (2 LDS buffers, next K-tile in flight during compute), with the
-// st_16x32 XOR swizzle (byte ^= ((byte>>9)&1)<<5 inside each 1024 B
-// subtile) applied to the *global source* address so the LDS image stays
-// lane-linear for glds; ds_read_b128 fragment reads apply the same XOR.
+// width 16 (2 LDS buffers, next K-tile in flight during compute), with a
+// conflict-free XOR swizzle (see swz() below) applied to the *global
+// source* address so the LDS image stays lane-linear for glds;
+// ds_read_b128 fragment reads apply the same XOR.
 // Partial tiles (M/N/K tails) take a bounds-checked vector-staging path
 // with an identical LDS image.
 #include "vfa_common.h"
@@ -50,8 +50,8 @@ __device__ __forceinline__ float act_f(float x, int kind) {
   return x;
 }
 
-// stage a (ROWS x 64) bf16 tile into an LDS image with the st_16x32
-// swizzle via glds; each wave covers (ROWS*128/1024)/WAVES subtiles.
+// stage a (ROWS x 64) bf16 tile into a swizzled LDS image (see swz())
+// via glds; each wave covers (ROWS*128/1024)/WAVES subtiles.
 // PIECES > 1 splits the wave's subtiles into issue groups so the glds for
 // the next K-tile can be spread across the current tile's MFMA quadrants
 // (piece = which group to issue; -1 = all).
