@@ -132,3 +132,19 @@ def test_conv2d_act_out_buffer_slice():
     ref = F.relu(F.conv2d(x, w, None, 1, 1))
     torch.testing.assert_close(buf[:, 8:32], ref)
     assert (buf[:, :8] == 0).all() and (buf[:, 32:] == 0).all()
+
+
+def test_conv2d_mod_padded_weight_cache_invalidates():
+    """The cached channel-padded weight must refresh when new weights are
+    loaded into the module (version-keyed like cached_cl_weight)."""
+    from video_features_amd.ops import conv2d_mod
+    torch.manual_seed(0)
+    conv = torch.nn.Conv2d(3, 16, 3, 1, 1)
+    x = torch.randn(1, 3, 8, 8)
+    y1 = conv2d_mod(conv, x)
+    with torch.no_grad():
+        conv.weight.mul_(2.0)
+    y2 = conv2d_mod(conv, x)
+    ref = torch.nn.functional.conv2d(x, conv.weight, conv.bias, 1, 1)
+    torch.testing.assert_close(y2, ref)
+    assert not torch.allclose(y1, y2)

@@ -504,33 +504,35 @@ def conv2d_mod(conv: torch.nn.Module, x: torch.Tensor, act: str = 'none',
                 and not _env_flag('VFA_FORCE_TORCH_OPS'))
     if eligible and w.shape[0] < 16 and res is None:
         # tiny-N heads (RAFT flow head N=2, per GRU iteration): pad the
-        # OUTPUT channels to 16 (cached), run in-tree, slice back
+        # OUTPUT channels to 16 (cached, weight-version-keyed), run
+        # in-tree, slice back
         kout = w.shape[0]
-        wp = getattr(conv, '_vfa_wnpad', None)
-        if wp is None or wp.device != x.device or wp.dtype != x.dtype:
+        key = (w._version, w.dtype, w.device, w.data_ptr())
+        ent = getattr(conv, '_vfa_wnpad', None)
+        if ent is None or ent[0] != key:
             wp = torch.nn.functional.pad(
                 w, (0, 0, 0, 0, 0, 0, 0, 16 - kout)).contiguous(
                     memory_format=torch.channels_last)
-            conv._vfa_wnpad = wp
-            conv._vfa_bnpad = (torch.nn.functional.pad(
-                conv.bias, (0, 16 - kout)) if conv.bias is not None
-                else None)
-        y = conv2d_act(x, wp, conv._vfa_bnpad, conv.stride, conv.padding,
-                       act)
+            bp = (torch.nn.functional.pad(conv.bias, (0, 16 - kout))
+                  if conv.bias is not None else None)
+            conv._vfa_wnpad = ent = (key, wp, bp)
+        y = conv2d_act(x, ent[1], ent[2], conv.stride, conv.padding, act)
         return y[:, :kout].contiguous(memory_format=torch.channels_last)
     if eligible and c % 8 != 0:
         # stems (C=3/2/1) and the RAFT corr input (C=324): zero-pad the
-        # channel dim — weight padded once and cached on the module, input
-        # padded inside the conv's (fused) pad pass
-        wp = getattr(conv, '_vfa_wpad', None)
-        if wp is None or wp.device != x.device or wp.dtype != x.dtype:
+        # channel dim — weight padded once and cached on the module (keyed
+        # on the weight VERSION so a later load_state_dict invalidates),
+        # input padded inside the conv's (fused) pad pass
+        key = (w._version, w.dtype, w.device, w.data_ptr())
+        ent = getattr(conv, '_vfa_wpad', None)
+        if ent is None or ent[0] != key:
             c8 = (c + 7) // 8 * 8
             wp = torch.nn.functional.pad(
                 w, (0, 0, 0, 0, 0, c8 - c)).contiguous(
                     memory_format=torch.channels_last)
-            conv._vfa_wpad = wp
-        return conv2d_act(x, wp, conv.bias, conv.stride, conv.padding, act,
-                          res)
+            conv._vfa_wpad = ent = (key, wp)
+        return conv2d_act(x, ent[1], conv.bias, conv.stride, conv.padding,
+                          act, res)
     if eligible and kh == 1 and kw == 1 and conv.stride == (1, 1) \
             and w.shape[0] % 8 == 0:
         return conv1x1_act(x, w, conv.bias, act, res)
