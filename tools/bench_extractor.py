@@ -29,6 +29,7 @@ def main():
     p.add_argument('--frames', type=int, default=120)
     p.add_argument('--size', type=int, default=224)
     p.add_argument('--extract_method', default='uni_12')
+    p.add_argument('--batch_size', type=int, default=32)
     p.add_argument('--cpu', action='store_true')
     args = p.parse_args()
 
@@ -44,9 +45,10 @@ def main():
 
     dev = torch.device('cpu' if args.cpu or not torch.cuda.is_available()
                        else 'cuda:0')
-    cfg = Config(feature_type=args.feature_type, video_paths=paths,
+    cfg = Config(feature_type=args.feature_type,
+                 batch_size=args.batch_size, video_paths=paths,
                  cpu=dev.type == 'cpu', extract_method=args.extract_method,
-                 batch_size=64, profile=True, tmp_path=os.path.join(tmp, 't'))
+                 profile=True, tmp_path=os.path.join(tmp, 't'))
     ex = get_extractor_class(cfg.feature_type)(cfg, external_call=True)
     ex.models_for(dev)                       # build outside the timing
     idxs = torch.arange(len(paths), device=dev)
