@@ -31,25 +31,30 @@ from ..runtime.sinks import (action_on_extraction, make_output_path,
 from ..runtime.progress import make_progress
 
 
-def decode_ahead(read_chunk, chunk_specs):
-    """Iterate ``(spec, read_chunk(spec))`` with the NEXT chunk's decode
-    running on a worker thread while the caller runs the current chunk on
-    the GPU — intra-video decode/compute overlap for extractors that stream
-    a video in batches (ResNet/RAFT/PWC; the per-video ``prepare`` hook
-    covers the whole-video-decode families)."""
+def decode_ahead(read_chunk, chunk_specs, depth: int = None):
+    """Iterate ``(spec, read_chunk(spec))`` with the next ``depth`` chunks'
+    decodes running on worker threads while the caller runs the current
+    chunk on the GPU — intra-video decode/compute overlap for extractors
+    that stream a video in batches (ResNet/RAFT/PWC; the per-video
+    ``prepare`` hook covers the whole-video-decode families)."""
+    import os
     from concurrent.futures import ThreadPoolExecutor
+    if depth is None:
+        depth = max(1, int(os.environ.get('VFA_DECODE_AHEAD', '2')))
     chunk_specs = list(chunk_specs)
     if len(chunk_specs) <= 1:
         for spec in chunk_specs:
             yield spec, read_chunk(spec)
         return
-    with ThreadPoolExecutor(max_workers=1,
+    with ThreadPoolExecutor(max_workers=depth,
                             thread_name_prefix='vfa-chunk-decode') as pool:
-        fut = pool.submit(read_chunk, chunk_specs[0])
+        futs = {p: pool.submit(read_chunk, chunk_specs[p])
+                for p in range(min(depth, len(chunk_specs)))}
         for i, spec in enumerate(chunk_specs):
-            cur = fut.result()
-            if i + 1 < len(chunk_specs):
-                fut = pool.submit(read_chunk, chunk_specs[i + 1])
+            cur = futs.pop(i).result()
+            nxt = i + depth
+            if nxt < len(chunk_specs):
+                futs[nxt] = pool.submit(read_chunk, chunk_specs[nxt])
             yield spec, cur
 
 
