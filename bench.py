@@ -501,7 +501,7 @@ def main():
     p.add_argument('--frame-batch', type=int, default=384,
                    help='CLIP: frames per forward chunk')
     p.add_argument('--clips-per-step', type=int, default=None,
-                   help='clips per rank per step (default: 8 for i3d_raft, '
+                   help='clips per rank per step (default: 16 for i3d_raft, '
                         '128 for vggish_r21d — measured throughput knees)')
     p.add_argument('--raft-iters', type=int, default=20)
     p.add_argument('--r21d-depth', type=int, choices=[18, 34], default=34,
@@ -519,7 +519,7 @@ def main():
     args = p.parse_args()
 
     if args.clips_per_step is None:
-        args.clips_per_step = 128 if args.model == 'vggish_r21d' else 8
+        args.clips_per_step = 128 if args.model == 'vggish_r21d' else 16
 
     rank, local_rank, world = get_dist()
     if torch.cuda.is_available():
