@@ -93,11 +93,14 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
   __bf16* s_vt = s_k + KVBLK * LSTR;                        // [D][LSTR]
   __bf16* s_p = s_vt + D * LSTR;                            // [4][16][LSTR]
 
-  const int bh = blockIdx.x;
-  const int bi = bh / h_total, hi = bh % h_total;
   const int qbase = blockIdx.y * QBLK;
   const int lane = threadIdx.x & 63, wave = threadIdx.x >> 6;
   const int lo = lane & 15, hi4 = lane >> 4;
+  // grid-stride over (batch, head) pairs: ViT-sized problems have tiny
+  // per-pair work (N=50), so several pairs per block amortize the block
+  // launch/tail and keep the staging pipeline warm
+  for (int bh = blockIdx.x; bh < b_total * h_total; bh += gridDim.x) {
+  const int bi = bh / h_total, hi = bh % h_total;
 
   const long long row_stride = 3LL * h_total * D;
   const __bf16* q_g = qkv + ((long long)bi * n + qbase) * row_stride +
@@ -236,6 +239,8 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
     for (int nt = 0; nt < 4; ++nt)
       orow[nt * 16 + lo] = (__bf16)(o_acc[nt][r] * inv_l);
   }
+  __syncthreads();   // LDS K/V reuse by the next (b, h) pair
+  }
 }
 
 // layout probe: D(16x16) = A(16x32) @ B(32x16), row-major f32 in/out
@@ -259,7 +264,8 @@ extern "C" {
 
 void vfa_flash_qkv(const void* qkv, void* out, int b, int n, int h,
                    float scale, hipStream_t stream) {
-  dim3 grid(b * h, (n + QBLK - 1) / QBLK);
+  // cap grid.x: blocks grid-stride over (b, h) pairs (4 blocks/CU fit)
+  dim3 grid(min(b * h, 2048), (n + QBLK - 1) / QBLK);
   size_t lds = (size_t)(QBLK + KVBLK + D + 4 * 16) * LSTR * sizeof(__bf16);
   hipLaunchKernelGGL(flash_qkv_kernel, grid, dim3(256), lds, stream,
                      (const __bf16*)qkv, (__bf16*)out, b, n, h, scale);
