@@ -116,3 +116,25 @@ def test_temporal_shard_partition(n, bs, tp):
             assert merged[j * tp + r] is None
             merged[j * tp + r] = s
     assert merged == starts
+
+
+def test_conv2d_act_routing_fuzz():
+    """conv2d_act must agree with F.conv2d across the routing space
+    (kernel sizes, strides, pads, channel alignments) on CPU — the same
+    dispatch logic picks kernels on GPU."""
+    import torch
+    import torch.nn.functional as F
+    from video_features_amd import ops
+    torch.manual_seed(0)
+    cases = [
+        (3, 8, 3, 3, 1, 1), (5, 16, 24, 1, 2, 0), (2, 12, 16, 5, 1, 2),
+        (1, 7, 16, 3, 1, 1), (2, 16, 20, 3, 2, 1), (1, 16, 16, 7, 2, 3),
+        (2, 24, 18, 1, 1, 0),
+    ]
+    for b, cin, cout, k, s, p in cases:
+        x = torch.randn(b, cin, 14, 15)
+        w = torch.randn(cout, cin, k, k)
+        bias = torch.randn(cout)
+        y = ops.conv2d_act(x, w, bias, s, p, 'relu')
+        ref = F.relu(F.conv2d(x, w, bias, s, p))
+        torch.testing.assert_close(y, ref, rtol=1e-4, atol=1e-4), (b, cin)
