@@ -142,7 +142,12 @@ class ResNet(nn.Module):
         x = ops.conv2d_mod(self.conv1, x, 'relu' if bn1_id else 'none')
         if not bn1_id:
             x = self.relu(self.bn1(x))
-        x = self.maxpool(x)
+        if (x.is_cuda and ops.hip_available()
+                and x.is_contiguous(memory_format=torch.channels_last)):
+            # our pool kernel skips the argmax indices torch always writes
+            x = ops.maxpool2d(x, (3, 3), (2, 2), 1, nhwc=True)
+        else:
+            x = self.maxpool(x)
         x = self.layer4(self.layer3(self.layer2(self.layer1(x))))
         return self.avgpool(x).flatten(1)
 

@@ -355,6 +355,24 @@ def maxpool2d_same(x: torch.Tensor, kernel, stride,
     return torch.nn.functional.max_pool2d(xp, tuple(kernel), tuple(stride))
 
 
+def maxpool2d(x: torch.Tensor, kernel, stride, padding,
+              nhwc: bool = False) -> torch.Tensor:
+    """torch-semantics max_pool2d (symmetric padding, clamped windows —
+    identical to torch's -inf padding) on the same HIP kernel; replaces
+    torch's NHWC maxpool (which always computes argmax indices) for the
+    ResNet stem."""
+    if _use_hip(x):
+        h, w = x.shape[-2:]
+        if nhwc and not x.is_contiguous(memory_format=torch.channels_last):
+            x = x.contiguous(memory_format=torch.channels_last)
+        out_sz = [(h + 2 * padding - kernel[0]) // stride[0] + 1,
+                  (w + 2 * padding - kernel[1]) // stride[1] + 1]
+        return _ext.maxpool2d_same(x, list(kernel), list(stride),
+                                   [padding, padding], out_sz, nhwc)
+    return torch.nn.functional.max_pool2d(x, tuple(kernel), tuple(stride),
+                                          padding)
+
+
 _ACT_IDS = {'none': 0, 'relu': 1, 'quick_gelu': 2, 'gelu': 3,
             'leaky_relu': 4}
 
