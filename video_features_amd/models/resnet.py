@@ -143,8 +143,11 @@ class ResNet(nn.Module):
         if not bn1_id:
             x = self.relu(self.bn1(x))
         if (x.is_cuda and ops.hip_available()
-                and x.is_contiguous(memory_format=torch.channels_last)):
-            # our pool kernel skips the argmax indices torch always writes
+                and x.is_contiguous(memory_format=torch.channels_last)
+                and ops._env_flag('VFA_STEMPOOL')):
+            # opt-in: the in-tree pool kernel measured ~3% SLOWER end-to-end
+            # than torch's NHWC maxpool at this shape (64ch 112x112; same-box
+            # A/B 32.3k vs 33.3k f/s), so torch stays the default here
             x = ops.maxpool2d(x, (3, 3), (2, 2), 1, nhwc=True)
         else:
             x = self.maxpool(x)
