@@ -243,6 +243,187 @@ void flash_qkv_kernel(const __bf16* __restrict__ qkv,   // (B,N,3,H,D)
   }
 }
 
+// ViT-length fast path (n <= KVBLK): one KV tile, one Q block.  The
+// general kernel above exposes the full K/V staging latency on every
+// (b, h) pair — global loads, barrier, ~16 MFMAs, repeat.  Here the
+// persistent loop is software-pipelined: while a pair is computed from
+// LDS buffer `buf`, the NEXT pair's K/V (2 uint4 each per thread) and
+// this wave's next Q A-fragments are already in flight to registers;
+// after the epilogue the registers drain into buffer `buf^1` and ONE
+// barrier publishes it.  K/V staging latency rides under the whole
+// softmax+MFMA body instead of serializing with it.
+__global__ __launch_bounds__(256)
+void flash_qkv_small_kernel(const __bf16* __restrict__ qkv,  // (B,N,3,H,D)
+                            __bf16* __restrict__ out,        // (B,N,H*D)
+                            int b_total, int n, int h_total, float scale) {
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  __bf16* s_k0 = reinterpret_cast<__bf16*>(smem);        // [2][KVBLK][LSTR]
+  __bf16* s_vt0 = s_k0 + 2 * KVBLK * LSTR;               // [2][D][LSTR]
+  __bf16* s_p = s_vt0 + 2 * D * LSTR;                    // [4][16][LSTR]
+
+  const int tid = threadIdx.x;
+  const int lane = tid & 63, wave = tid >> 6;
+  const int lo = lane & 15, hi4 = lane >> 4;
+  const long long row_stride = 3LL * h_total * D;
+  const int total = b_total * h_total;
+  const int qr = min(wave * 16 + lo, n - 1);   // this wave's A-frag row
+
+  int bh = blockIdx.x;
+  if (bh >= total) return;
+
+  // prologue: stage pair bh into buffer 0, prefetch its Q fragments
+  bf16x8 a_cur[2];
+  {
+    const int bi = bh / h_total, hi = bh % h_total;
+    const __bf16* k_g = qkv + (long long)bi * n * row_stride +
+                        ((long long)1 * h_total + hi) * D;
+    stage_rows(k_g, KVBLK, row_stride, s_k0, n);
+    stage_vt(k_g + (long long)h_total * D, row_stride, s_vt0, n);
+    const __bf16* q_g = qkv + (long long)bi * n * row_stride +
+                        (long long)hi * D;
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt)
+      a_cur[kt] = *reinterpret_cast<const bf16x8*>(
+          q_g + (long long)qr * row_stride + kt * 32 + hi4 * 8);
+  }
+  __syncthreads();
+
+  int buf = 0;
+  for (; bh < total; bh += gridDim.x) {
+    const int nbh = bh + gridDim.x;
+    const bool has_next = nbh < total;
+    // ---- issue next pair's loads (registers; drained after compute).
+    // A tail block without a next pair re-reads its own pair: the loads
+    // are discarded, but every thread still walks the same code.
+    const int pf = has_next ? nbh : bh;
+    const int pbi = pf / h_total, phi = pf % h_total;
+    const __bf16* k_gn = qkv + (long long)pbi * n * row_stride +
+                         ((long long)1 * h_total + phi) * D;
+    const __bf16* v_gn = k_gn + (long long)h_total * D;
+    uint4 kreg[2], vreg[2];
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int t = tid + i * 256, row = t >> 3, seg = t & 7;
+      kreg[i] = row < n ? *reinterpret_cast<const uint4*>(
+                              k_gn + row * row_stride + seg * 8)
+                        : uint4{0u, 0u, 0u, 0u};
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int t = tid + i * 256, key = t & 63, seg = t >> 6;
+      vreg[i] = key < n ? *reinterpret_cast<const uint4*>(
+                              v_gn + key * row_stride + seg * 8)
+                        : uint4{0u, 0u, 0u, 0u};
+    }
+    bf16x8 a_next[2];
+    {
+      const __bf16* q_gn = qkv + (long long)pbi * n * row_stride +
+                           (long long)phi * D;
+#pragma unroll
+      for (int kt = 0; kt < 2; ++kt)
+        a_next[kt] = *reinterpret_cast<const bf16x8*>(
+            q_gn + (long long)qr * row_stride + kt * 32 + hi4 * 8);
+    }
+
+    // ---- compute current pair from LDS buffer `buf`
+    const __bf16* s_k = s_k0 + buf * KVBLK * LSTR;
+    const __bf16* s_vt = s_vt0 + buf * D * LSTR;
+    f32x4 s_frag[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) s_frag[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt)
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            s_k + (nt * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
+        s_frag[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a_cur[kt], bfr, s_frag[nt], 0, 0, 0);
+      }
+
+    float p_vals[4][4], l_run[4];
+#pragma unroll
+    for (int r = 0; r < 4; ++r) {
+      float mx = -1e30f;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        float s = s_frag[nt][r] * scale;
+        if (nt * 16 + lo >= n) s = -1e30f;
+        s_frag[nt][r] = s;
+        mx = fmaxf(mx, s);
+      }
+      mx = group16_max(mx);
+      float rowsum = 0.f;
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        const float p = __expf(s_frag[nt][r] - mx);
+        p_vals[nt][r] = p;
+        rowsum += p;
+      }
+      l_run[r] = group16_sum(rowsum);
+    }
+
+    __bf16* p_lds = s_p + wave * 16 * LSTR;
+#pragma unroll
+    for (int nt = 0; nt < 4; ++nt)
+#pragma unroll
+      for (int r = 0; r < 4; ++r)
+        p_lds[(hi4 * 4 + r) * LSTR + nt * 16 + lo] = (__bf16)p_vals[nt][r];
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+    f32x4 o_acc[4];
+#pragma unroll
+    for (int t = 0; t < 4; ++t) o_acc[t] = f32x4{0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+    for (int kt = 0; kt < 2; ++kt) {
+      bf16x8 a = *reinterpret_cast<const bf16x8*>(
+          p_lds + lo * LSTR + kt * 32 + hi4 * 8);
+#pragma unroll
+      for (int nt = 0; nt < 4; ++nt) {
+        bf16x8 bfr = *reinterpret_cast<const bf16x8*>(
+            s_vt + (nt * 16 + lo) * LSTR + kt * 32 + hi4 * 8);
+        o_acc[nt] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+            a, bfr, o_acc[nt], 0, 0, 0);
+      }
+    }
+
+    {
+      const int bi = bh / h_total, hi = bh % h_total;
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int grow = wave * 16 + hi4 * 4 + r;
+        if (grow >= n) continue;
+        const float inv_l = 1.0f / l_run[r];
+        __bf16* orow = out + ((long long)bi * n + grow) * (h_total * D) +
+                       (long long)hi * D;
+#pragma unroll
+        for (int nt = 0; nt < 4; ++nt)
+          orow[nt * 16 + lo] = (__bf16)(o_acc[nt][r] * inv_l);
+      }
+    }
+
+    // ---- drain the prefetched pair into the other buffer, ONE barrier
+    __bf16* d_k = s_k0 + (buf ^ 1) * KVBLK * LSTR;
+    __bf16* d_vt = s_vt0 + (buf ^ 1) * D * LSTR;
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int t = tid + i * 256, row = t >> 3, seg = t & 7;
+      *reinterpret_cast<uint4*>(d_k + row * LSTR + seg * 8) = kreg[i];
+    }
+#pragma unroll
+    for (int i = 0; i < 2; ++i) {
+      const int t = tid + i * 256, key = t & 63, seg = t >> 6;
+      const __bf16* e = reinterpret_cast<const __bf16*>(&vreg[i]);
+#pragma unroll
+      for (int j = 0; j < 8; ++j) d_vt[(seg * 8 + j) * LSTR + key] = e[j];
+    }
+    __syncthreads();
+    buf ^= 1;
+    a_cur[0] = a_next[0];
+    a_cur[1] = a_next[1];
+  }
+}
+
 // layout probe: D(16x16) = A(16x32) @ B(32x16), row-major f32 in/out
 __global__ void mfma_gemm16_kernel(const float* a, const float* b, float* d) {
   const int l = threadIdx.x, lo = l & 15, hi4 = l >> 4;
@@ -264,6 +445,21 @@ extern "C" {
 
 void vfa_flash_qkv(const void* qkv, void* out, int b, int n, int h,
                    float scale, hipStream_t stream) {
+  static const bool no_pipe = [] {
+    const char* e = getenv("VFA_NO_FLASHPIPE");
+    return e && e[0] && e[0] != '0';
+  }();
+  if (n <= KVBLK && !no_pipe) {
+    // ViT-length fast path: persistent blocks, double-buffered K/V,
+    // next-pair prefetch under the current pair's compute.  Cap the grid
+    // so each block iterates >= 2 pairs when the problem allows — the
+    // pipeline only pays when there IS a next pair.
+    dim3 grid(min(b * h, 1024), 1);
+    size_t lds = (size_t)(2 * (KVBLK + D) + 4 * 16) * LSTR * sizeof(__bf16);
+    hipLaunchKernelGGL(flash_qkv_small_kernel, grid, dim3(256), lds, stream,
+                       (const __bf16*)qkv, (__bf16*)out, b, n, h, scale);
+    return;
+  }
   // cap grid.x: blocks grid-stride over (b, h) pairs (4 blocks/CU fit)
   dim3 grid(min(b * h, 2048), (n + QBLK - 1) / QBLK);
   size_t lds = (size_t)(QBLK + KVBLK + D + 4 * 16) * LSTR * sizeof(__bf16);
