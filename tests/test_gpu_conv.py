@@ -131,3 +131,29 @@ def test_pad_matches_torch():
         .contiguous(memory_format=torch.channels_last)
     # exercised through conv2d_act pad path vs F.conv2d's implicit pad
     _run(2, 16, 9, 11, 24, 3, 3, 1, (1, 1))
+
+
+# ---- ResNet bottleneck DOWNSAMPLE 1x1 convs (stride 2 at the stage
+# transitions, stride 1 at layer1.0) — routed in-tree by
+# models/resnet.py via conv2d_mod (reference pulls these from
+# torchvision, models/resnet/extract_resnet.py:54-67)
+@pytest.mark.parametrize('c,hw,k,stride', [
+    (64, 56, 256, 1), (256, 56, 512, 2), (512, 28, 1024, 2),
+    (1024, 14, 2048, 2),
+])
+def test_conv1x1_downsample_shapes(c, hw, k, stride):
+    _run(4, c, hw, hw, k, 1, 1, stride, (0, 0))
+
+
+def test_conv2d_mod_downsample_route():
+    """conv2d_mod on a folded downsample [conv, Identity] matches the
+    eager nn.Sequential on GPU (the Bottleneck fused path's idt)."""
+    _hip()
+    torch.manual_seed(0)
+    dev = 'cuda:0'
+    conv = torch.nn.Conv2d(256, 512, 1, 2, 0).to(dev, torch.bfloat16)
+    x = (torch.randn(2, 256, 56, 56, device=dev) * 0.5) \
+        .to(torch.bfloat16).contiguous(memory_format=torch.channels_last)
+    out = ops.conv2d_act(x, conv.weight, conv.bias, 2, 0, 'none')
+    ref = F.conv2d(x.float(), conv.weight.float(), conv.bias.float(), 2, 0)
+    assert ((out.float() - ref).abs() / ref.abs().max()).max().item() < 0.08

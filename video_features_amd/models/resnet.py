@@ -42,8 +42,11 @@ class BasicBlock(nn.Module):
                 and x.is_contiguous(memory_format=torch.channels_last)
                 and ops.hip_available()):
             # both 3x3 convs through the in-tree implicit-GEMM kernel;
-            # conv2 fuses the residual add + ReLU into its epilogue
-            idt = x if self.downsample is None else self.downsample(x)
+            # conv2 fuses the residual add + ReLU into its epilogue.
+            # The downsample 1x1 (BN folded -> [conv, Identity]) also goes
+            # in-tree: nn.Conv2d.forward would hand it to MIOpen.
+            idt = x if self.downsample is None else \
+                ops.conv2d_mod(self.downsample[0], x, 'none')
             if not idt.is_contiguous(memory_format=torch.channels_last):
                 idt = idt.contiguous(memory_format=torch.channels_last)
             y = ops.conv2d_act(x, self.conv1.weight, self.conv1.bias,
@@ -89,8 +92,13 @@ class Bottleneck(nn.Module):
             # 1x1 convs as fused MFMA GEMMs on the CL view: conv1+ReLU in
             # one epilogue; conv3 + residual-add + ReLU in one epilogue
             # (removes MIOpen's SubTensor zero-fill + the eager add/relu
-            # round trips — see profiles/)
-            idt = x if self.downsample is None else self.downsample(x)
+            # round trips — see profiles/).  The downsample 1x1 (stride 1
+            # or 2; BN folded -> [conv, Identity]) routes in-tree too:
+            # left as self.downsample(x) it was ~12% of steady-state
+            # kernel time on MIOpen igemm + SubTensor zero-fill
+            # (profiles/resnet50_profile_final_r02.md).
+            idt = x if self.downsample is None else \
+                ops.conv2d_mod(self.downsample[0], x, 'none')
             if self.conv1.out_channels >= 128:
                 y = ops.conv1x1_act(x, self.conv1.weight, self.conv1.bias,
                                     'relu')
