@@ -99,12 +99,8 @@ class Bottleneck(nn.Module):
             # (profiles/resnet50_profile_final_r02.md).
             idt = x if self.downsample is None else \
                 ops.conv2d_mod(self.downsample[0], x, 'none')
-            if self.conv1.out_channels >= 128:
-                y = ops.conv1x1_act(x, self.conv1.weight, self.conv1.bias,
-                                    'relu')
-            else:
-                y = F.relu(F.conv2d(x, self.conv1.weight, self.conv1.bias),
-                           inplace=True)
+            y = ops.conv1x1_act(x, self.conv1.weight, self.conv1.bias,
+                                'relu')
             # 3x3 conv + bias + ReLU as ONE in-tree implicit-GEMM MFMA
             # kernel (conv2d.hip — 1.4-1.5x MIOpen on these shapes,
             # gpurun_out/bench_conv_r2b.log)
