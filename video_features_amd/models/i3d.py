@@ -297,5 +297,10 @@ class I3D(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = self.dropout(self._backbone(x))
-        logits = self.conv3d_0c_1x1(x)          # (B, K, t, 1, 1)
-        return logits.mean(dim=(2, 3, 4))       # time-averaged class scores
+        # the 1x1x1 logits conv over (B, 1024, t, 1, 1) as a plain linear
+        # on the channel dim — keeps the model conv3d-free end to end
+        u = self.conv3d_0c_1x1
+        w = u.conv.weight.reshape(u.conv.weight.shape[0], -1)   # (K, 1024)
+        logits = F.linear(x.squeeze(-1).squeeze(-1).transpose(1, 2), w,
+                          u.conv.bias)           # (B, t, K)
+        return logits.mean(dim=1)                # time-averaged class scores
