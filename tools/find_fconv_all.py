@@ -20,7 +20,7 @@ def run(tag, fn):
         fn()
     torch.cuda.synchronize()
     print(tag, 'leaks:', len(leaks))
-    for l in leaks[:8]:
+    for l in leaks[:40]:
         print('  ', l)
 
 # PWC at bench shape
