@@ -253,3 +253,20 @@ def test_i3d_show_pred_prints_kinetics_top5(tmp_path, capsys):
     assert sum(1 for line in out.splitlines() if 'prob' in line
                or '%' in line or '\t' in line) >= 5 or 'top' in out.lower() \
         or out.count('\n') >= 6
+
+
+def test_i3d_logits_head_is_exact_linear():
+    """forward()'s channel-dim linear logits head equals the reference's
+    1x1x1 conv3d head (reference i3d_net.py:238-264) applied explicitly —
+    the model stays conv3d-free without changing show_pred scores."""
+    import torch
+    from video_features_amd.models.i3d import I3D
+    torch.manual_seed(0)
+    m = I3D(num_classes=7).eval()
+    x = torch.randn(2, 1024, 5, 1, 1)
+    u = m.conv3d_0c_1x1
+    ref = u.conv(x).mean(dim=(2, 3, 4))       # explicit conv3d head
+    w = u.conv.weight.reshape(u.conv.weight.shape[0], -1)
+    got = torch.nn.functional.linear(
+        x.squeeze(-1).squeeze(-1).transpose(1, 2), w, u.conv.bias).mean(dim=1)
+    assert torch.allclose(got, ref, atol=1e-5), (got - ref).abs().max()
