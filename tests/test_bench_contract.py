@@ -31,6 +31,11 @@ def test_bench_single_rank_contract():
     assert d['n_gpus'] == 1 and d['steps'] == 2 and d['warmup'] == 1
     assert d['data'] == 'synthetic' and d['scaling'] == 'weak'
     assert d['value'] > 0
+    # the default (flagless) run must measure BASELINE.json's headline
+    # metric on its named config
+    assert d['metric'] == 'frames/sec CLIP-ViT-B/32 uni_12'
+    assert d['config']['model'] == 'CLIP-ViT-B/32'
+    assert d['config']['parallelism'] == 'dp1'
 
 
 def test_bench_two_rank_gloo():
