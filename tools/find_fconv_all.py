@@ -50,9 +50,13 @@ i3 = I3D(modality='rgb').eval().to(dev, dt); fold_batchnorms(i3)
 xi = torch.rand(2, 3, 64, 224, 224, device=dev).to(dt)
 run('i3d', lambda: i3(xi))
 
-# RAFT
+# RAFT — feed channels_last as the extractor/bench do (use_channels_last
+# flips the module; an NCHW input here reports every conv as a false leak)
 from video_features_amd.models.raft import RAFT
 rf = RAFT().eval().to(dev, dt); fold_batchnorms(rf)
-f1 = torch.rand(8, 3, 224, 224, device=dev).to(dt) * 255
+if hasattr(rf, 'use_channels_last'):
+    rf = rf.use_channels_last()
+f1 = (torch.rand(8, 3, 224, 224, device=dev).to(dt) * 255).contiguous(
+    memory_format=torch.channels_last)
 run('raft', lambda: rf(f1, f1, iters=3))
 print('done')
